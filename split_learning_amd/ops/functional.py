@@ -1,0 +1,373 @@
+"""Autograd wrappers around the hand-written CDNA4 HIP kernels.
+
+Every Function here is GPU-only (callers dispatch CPU tensors to plain torch in
+split_learning_amd/ops/modules.py).  Numerics are exact fp32: the GEMM-shaped ops
+run on the f32-input MFMA path (v_mfma_f32_16x16x4_f32 — bitwise an fmaf chain),
+matching the reference's fp32 PyTorch math (reference trains pure fp32:
+src/train/VGG16.py has no dtype casts).
+
+Kernel-side entry points live in split_learning_amd/ops/csrc/ and are exported by
+the in-tree _sl_kernels extension (see ops/__init__.py loading policy).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+from torch.autograd import Function
+
+from . import native
+
+
+# ---------------------------------------------------------------------------
+# GEMM / Linear
+# ---------------------------------------------------------------------------
+
+def matmul_f32(a: torch.Tensor, b: torch.Tensor, trans_a: bool = False, trans_b: bool = False,
+               out: Optional[torch.Tensor] = None, accumulate: bool = False) -> torch.Tensor:
+    """C = op(A) @ op(B) for 2-D or batched 3-D fp32 tensors (strided MFMA GEMM)."""
+    return native().matmul_f32(a, b, trans_a, trans_b, out, accumulate)
+
+
+class LinearFn(Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        # x: [*, K] -> flatten rows; weight: [N, K]; y = x @ w^T + b
+        xs = x.reshape(-1, x.shape[-1])
+        y = native().linear_fwd(xs.contiguous(), weight, bias)
+        ctx.save_for_backward(xs, weight)
+        ctx.has_bias = bias is not None
+        ctx.x_shape = x.shape
+        return y.reshape(*x.shape[:-1], weight.shape[0])
+
+    @staticmethod
+    def backward(ctx, gy):
+        xs, weight = ctx.saved_tensors
+        gys = gy.reshape(-1, gy.shape[-1]).contiguous()
+        gx = gw = gb = None
+        if ctx.needs_input_grad[0]:
+            gx = matmul_f32(gys, weight)               # [M,N]x[N,K] -> [M,K]
+            gx = gx.reshape(ctx.x_shape)
+        if ctx.needs_input_grad[1]:
+            gw = matmul_f32(gys, xs, trans_a=True)     # [N,M]x[M,K] -> [N,K]
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            gb = native().colsum_f32(gys)
+        return gx, gw, gb
+
+
+def linear(x, weight, bias=None):
+    return LinearFn.apply(x, weight, bias)
+
+
+# ---------------------------------------------------------------------------
+# Convolution (implicit-GEMM, NCHW)
+# ---------------------------------------------------------------------------
+
+class Conv2dFn(Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, padding):
+        x = x.contiguous()
+        y = native().conv2d_fwd(x, weight, bias, stride, padding)
+        ctx.save_for_backward(x, weight)
+        ctx.stride = stride
+        ctx.padding = padding
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, weight = ctx.saved_tensors
+        gy = gy.contiguous()
+        gx = gw = gb = None
+        if ctx.needs_input_grad[0]:
+            gx = native().conv2d_bwd_data(gy, weight, ctx.stride, ctx.padding,
+                                          x.shape[2], x.shape[3])
+        if ctx.needs_input_grad[1]:
+            gw = native().conv2d_bwd_weight(gy, x, weight.shape[2], weight.shape[3],
+                                            ctx.stride, ctx.padding)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            gb = native().conv2d_bwd_bias(gy)
+        return gx, gw, gb, None, None
+
+
+def conv2d(x, weight, bias=None, stride=1, padding=0):
+    return Conv2dFn.apply(x, weight, bias, int(stride), int(padding))
+
+
+# ---------------------------------------------------------------------------
+# BatchNorm2d (training + eval), optional fused ReLU
+# ---------------------------------------------------------------------------
+
+class BatchNorm2dFn(Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, training, momentum, eps,
+                fuse_relu):
+        x = x.contiguous()
+        if training:
+            mean, var = native().bn2d_stats(x)  # biased var over (N,H,W)
+            invstd = (var + eps).rsqrt()
+            # running-stat update matches torch: unbiased var for running_var
+            with torch.no_grad():
+                n = x.numel() / x.shape[1]
+                unbiased = var * (n / max(n - 1.0, 1.0))
+                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+        else:
+            mean = running_mean
+            invstd = (running_var + eps).rsqrt()
+        y = native().bn2d_fwd(x, mean, invstd, gamma, beta, fuse_relu)
+        ctx.save_for_backward(x, gamma, mean, invstd, y)
+        ctx.training = training
+        ctx.fuse_relu = fuse_relu
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, gamma, mean, invstd, y = ctx.saved_tensors
+        gy = gy.contiguous()
+        if ctx.fuse_relu:
+            gy = native().relu_bwd(gy, y)
+        if ctx.training:
+            gx, ggamma, gbeta = native().bn2d_bwd(x, gy, gamma, mean, invstd)
+        else:
+            # eval-mode backward: per-channel affine with fixed stats
+            gx, ggamma, gbeta = native().bn2d_bwd_eval(x, gy, gamma, mean, invstd)
+        return gx, ggamma, gbeta, None, None, None, None, None, None
+
+
+def batch_norm2d(x, gamma, beta, running_mean, running_var, training, momentum=0.1,
+                 eps=1e-5, fuse_relu=False):
+    return BatchNorm2dFn.apply(x, gamma, beta, running_mean, running_var, training,
+                               momentum, eps, fuse_relu)
+
+
+# ---------------------------------------------------------------------------
+# ReLU / GELU / Tanh
+# ---------------------------------------------------------------------------
+
+class ReluFn(Function):
+    @staticmethod
+    def forward(ctx, x):
+        y = native().relu_fwd(x.contiguous())
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        (y,) = ctx.saved_tensors
+        return native().relu_bwd(gy.contiguous(), y)
+
+
+def relu(x):
+    return ReluFn.apply(x)
+
+
+class GeluFn(Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        y = native().gelu_fwd(x)  # exact erf GELU (matches nn.GELU default)
+        ctx.save_for_backward(x)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        (x,) = ctx.saved_tensors
+        return native().gelu_bwd(gy.contiguous(), x)
+
+
+def gelu(x):
+    return GeluFn.apply(x)
+
+
+class TanhFn(Function):
+    @staticmethod
+    def forward(ctx, x):
+        y = native().tanh_fwd(x.contiguous())
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        (y,) = ctx.saved_tensors
+        return native().tanh_bwd(gy.contiguous(), y)
+
+
+def tanh(x):
+    return TanhFn.apply(x)
+
+
+# ---------------------------------------------------------------------------
+# MaxPool2d 2x2 stride 2 (the only pooling the model zoo uses)
+# ---------------------------------------------------------------------------
+
+class MaxPool2x2Fn(Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        y, idx = native().maxpool2x2_fwd(x)
+        ctx.save_for_backward(idx)
+        ctx.in_hw = (x.shape[2], x.shape[3])
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        (idx,) = ctx.saved_tensors
+        return native().maxpool2x2_bwd(gy.contiguous(), idx, ctx.in_hw[0], ctx.in_hw[1])
+
+
+def maxpool2x2(x):
+    return MaxPool2x2Fn.apply(x)
+
+
+# ---------------------------------------------------------------------------
+# Dropout (philox-style counter hash; mask stashed for backward)
+# ---------------------------------------------------------------------------
+
+class DropoutFn(Function):
+    @staticmethod
+    def forward(ctx, x, p, seed, offset):
+        x = x.contiguous()
+        y, mask = native().dropout_fwd(x, float(p), int(seed), int(offset))
+        ctx.save_for_backward(mask)
+        ctx.p = float(p)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        (mask,) = ctx.saved_tensors
+        return native().dropout_bwd(gy.contiguous(), mask, ctx.p), None, None, None
+
+
+_DROPOUT_STATE = {"seed": 0x5EEDC0DE, "offset": 0}
+
+
+def seed_dropout(seed: int) -> None:
+    _DROPOUT_STATE["seed"] = int(seed) & 0xFFFFFFFFFFFFFFFF
+    _DROPOUT_STATE["offset"] = 0
+
+
+def dropout(x, p, training=True):
+    if not training or p == 0.0:
+        return x
+    _DROPOUT_STATE["offset"] += 1
+    return DropoutFn.apply(x, p, _DROPOUT_STATE["seed"], _DROPOUT_STATE["offset"])
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm
+# ---------------------------------------------------------------------------
+
+class LayerNormFn(Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps):
+        xs = x.reshape(-1, x.shape[-1]).contiguous()
+        y, mean, invstd = native().layernorm_fwd(xs, gamma, beta, float(eps))
+        ctx.save_for_backward(xs, gamma, mean, invstd)
+        ctx.x_shape = x.shape
+        return y.reshape(x.shape)
+
+    @staticmethod
+    def backward(ctx, gy):
+        xs, gamma, mean, invstd = ctx.saved_tensors
+        gys = gy.reshape(-1, gy.shape[-1]).contiguous()
+        gx, ggamma, gbeta = native().layernorm_bwd(gys, xs, gamma, mean, invstd)
+        return gx.reshape(ctx.x_shape), ggamma, gbeta, None
+
+
+def layer_norm(x, gamma, beta, eps=1e-5):
+    return LayerNormFn.apply(x, gamma, beta, eps)
+
+
+# ---------------------------------------------------------------------------
+# Softmax (last dim) — used by attention
+# ---------------------------------------------------------------------------
+
+class SoftmaxFn(Function):
+    @staticmethod
+    def forward(ctx, x):
+        xs = x.reshape(-1, x.shape[-1]).contiguous()
+        y = native().softmax_fwd(xs)
+        ctx.save_for_backward(y)
+        ctx.x_shape = x.shape
+        return y.reshape(x.shape)
+
+    @staticmethod
+    def backward(ctx, gy):
+        (y,) = ctx.saved_tensors
+        gys = gy.reshape(-1, gy.shape[-1]).contiguous()
+        return native().softmax_bwd(gys, y).reshape(ctx.x_shape)
+
+
+def softmax_lastdim(x):
+    return SoftmaxFn.apply(x)
+
+
+# ---------------------------------------------------------------------------
+# Cross-entropy (fused log-softmax + NLL + NaN detect, mean reduction)
+# ---------------------------------------------------------------------------
+
+class CrossEntropyFn(Function):
+    @staticmethod
+    def forward(ctx, logits, labels):
+        logits = logits.contiguous()
+        loss, probs = native().ce_fwd(logits, labels)
+        ctx.save_for_backward(probs, labels)
+        return loss
+
+    @staticmethod
+    def backward(ctx, gloss):
+        probs, labels = ctx.saved_tensors
+        return native().ce_bwd(probs, labels, gloss), None
+
+
+def cross_entropy(logits, labels):
+    return CrossEntropyFn.apply(logits, labels)
+
+
+# ---------------------------------------------------------------------------
+# Embedding
+# ---------------------------------------------------------------------------
+
+class EmbeddingFn(Function):
+    @staticmethod
+    def forward(ctx, ids, weight, padding_idx):
+        ids = ids.contiguous()
+        y = native().embedding_fwd(ids, weight)
+        ctx.save_for_backward(ids)
+        ctx.num_embeddings = weight.shape[0]
+        ctx.dim = weight.shape[1]
+        ctx.padding_idx = -1 if padding_idx is None else int(padding_idx)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        (ids,) = ctx.saved_tensors
+        gw = native().embedding_bwd(ids, gy.contiguous(), ctx.num_embeddings,
+                                    ctx.padding_idx)
+        return None, gw, None
+
+
+def embedding(ids, weight, padding_idx=None):
+    return EmbeddingFn.apply(ids, weight, padding_idx)
+
+
+# ---------------------------------------------------------------------------
+# Fused optimizers (multi-tensor)
+# ---------------------------------------------------------------------------
+
+def sgd_step(params, grads, momentum_bufs, lr, momentum, weight_decay=0.0,
+             first_step=False):
+    """Fused multi-tensor SGD+momentum matching torch.optim.SGD semantics
+    (buf = m*buf + g; p -= lr*buf; first step: buf = g)."""
+    native().sgd_step(list(params), list(grads), list(momentum_bufs),
+                      float(lr), float(momentum), float(weight_decay), bool(first_step))
+
+
+def adamw_step(params, grads, exp_avgs, exp_avg_sqs, step, lr, beta1, beta2, eps,
+               weight_decay):
+    """Fused multi-tensor AdamW matching torch.optim.AdamW."""
+    native().adamw_step(list(params), list(grads), list(exp_avgs), list(exp_avg_sqs),
+                        int(step), float(lr), float(beta1), float(beta2), float(eps),
+                        float(weight_decay))

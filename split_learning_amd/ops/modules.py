@@ -1,0 +1,150 @@
+"""HIP-backed drop-in layer modules.
+
+Each class subclasses the corresponding torch.nn module so parameter/buffer names
+(and therefore state_dict keys and .pth files) are identical to the reference
+models (reference src/model/VGG16_CIFAR10.py etc.).  Dispatch rule:
+
+* input on GPU  -> hand-written CDNA4 kernels via ops.functional (REQUIRED —
+  raises if the native extension is missing; no silent eager fallback);
+* input on CPU  -> the stock torch forward (used by the CPU test suite and by
+  server-side validation on CPU-only hosts).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import functional as hf
+
+
+def _on_gpu(x: torch.Tensor) -> bool:
+    return x.is_cuda
+
+
+class HipConv2d(nn.Conv2d):
+    def forward(self, x):
+        if _on_gpu(x):
+            return hf.conv2d(x, self.weight, self.bias,
+                             stride=self.stride[0], padding=self.padding[0])
+        return super().forward(x)
+
+
+class HipLinear(nn.Linear):
+    def forward(self, x):
+        if _on_gpu(x):
+            return hf.linear(x, self.weight, self.bias)
+        return super().forward(x)
+
+
+class HipBatchNorm2d(nn.BatchNorm2d):
+    def forward(self, x):
+        if _on_gpu(x):
+            if self.training and self.track_running_stats and self.num_batches_tracked is not None:
+                self.num_batches_tracked.add_(1)
+            return hf.batch_norm2d(x, self.weight, self.bias, self.running_mean,
+                                   self.running_var, self.training, self.momentum,
+                                   self.eps, fuse_relu=False)
+        return super().forward(x)
+
+
+class HipReLU(nn.ReLU):
+    def forward(self, x):
+        if _on_gpu(x):
+            return hf.relu(x)
+        return super().forward(x)
+
+
+class HipGELU(nn.GELU):
+    def forward(self, x):
+        if _on_gpu(x):
+            return hf.gelu(x)
+        return super().forward(x)
+
+
+class HipTanh(nn.Tanh):
+    def forward(self, x):
+        if _on_gpu(x):
+            return hf.tanh(x)
+        return super().forward(x)
+
+
+class HipMaxPool2d(nn.MaxPool2d):
+    def forward(self, x):
+        if _on_gpu(x) and self.kernel_size == 2 and self.stride == 2:
+            return hf.maxpool2x2(x)
+        return super().forward(x)
+
+
+class HipDropout(nn.Dropout):
+    def forward(self, x):
+        if _on_gpu(x):
+            return hf.dropout(x, self.p, self.training)
+        return super().forward(x)
+
+
+class HipLayerNorm(nn.LayerNorm):
+    def forward(self, x):
+        if _on_gpu(x) and len(self.normalized_shape) == 1:
+            return hf.layer_norm(x, self.weight, self.bias, self.eps)
+        return super().forward(x)
+
+
+class HipEmbedding(nn.Embedding):
+    def forward(self, ids):
+        if ids.is_cuda:
+            return hf.embedding(ids, self.weight, self.padding_idx)
+        return super().forward(ids)
+
+
+class HipMultiheadAttention(nn.MultiheadAttention):
+    """Self-attention-only MHA (q=k=v, batch_first) backed by HIP GEMM+softmax.
+
+    Parameter layout (in_proj_weight [3E,E], in_proj_bias, out_proj.*) matches
+    nn.MultiheadAttention so KWT/ViT state dicts interoperate
+    (reference src/model/KWT_SPEECHCOMMANDS.py:9, other/.../ViT_CIFAR10.py:7).
+    """
+
+    def forward(self, query, key, value, **kwargs):  # type: ignore[override]
+        if not _on_gpu(query) or (query is not key) or (key is not value) or not self.batch_first:
+            return super().forward(query, key, value, **kwargs)
+        x = query  # [B, S, E]
+        B, S, E = x.shape
+        H = self.num_heads
+        hd = E // H
+        qkv = hf.linear(x, self.in_proj_weight, self.in_proj_bias)  # [B,S,3E]
+        q, k, v = qkv.split(E, dim=-1)
+        # [B,S,H,hd] -> [B*H, S, hd]
+        q = q.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd).contiguous()
+        k = k.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd).contiguous()
+        v = v.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd).contiguous()
+        scores = hf.matmul_f32(q, k, trans_b=True) * (1.0 / math.sqrt(hd))
+        probs = hf.softmax_lastdim(scores)
+        if self.dropout > 0.0:
+            probs = hf.dropout(probs, self.dropout, self.training)
+        ctxv = hf.matmul_f32(probs, v)  # [B*H, S, hd]
+        ctxv = ctxv.reshape(B, H, S, hd).permute(0, 2, 1, 3).reshape(B, S, E)
+        out = hf.linear(ctxv, self.out_proj.weight, self.out_proj.bias)
+        return out, None
+
+
+def attention_core(q, k, v, dropout_p: float = 0.0, training: bool = False,
+                   scale: Optional[float] = None):
+    """scaled-dot-product attention on [B*H, S, hd] tensors via HIP kernels."""
+    hd = q.shape[-1]
+    s = scale if scale is not None else 1.0 / math.sqrt(hd)
+    if q.is_cuda:
+        scores = hf.matmul_f32(q, k, trans_b=True) * s
+        probs = hf.softmax_lastdim(scores)
+        if dropout_p > 0.0:
+            probs = hf.dropout(probs, dropout_p, training)
+        return hf.matmul_f32(probs, v)
+    scores = torch.matmul(q, k.transpose(-1, -2)) * s
+    probs = F.softmax(scores, dim=-1)
+    if dropout_p > 0.0:
+        probs = F.dropout(probs, dropout_p, training)
+    return torch.matmul(probs, v)
