@@ -1,0 +1,57 @@
+// Common helpers for the split_learning_amd CDNA4 (gfx950) kernels.
+// All kernels here are written directly in HIP for MI355X — wave64, MFMA
+// matrix cores, LDS — no CUDA compatibility paths.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define SLK_WAVE 64
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+#define HIP_CHECK(expr)                                                          \
+  do {                                                                           \
+    hipError_t _e = (expr);                                                      \
+    if (_e != hipSuccess) {                                                      \
+      TORCH_CHECK(false, "HIP error: ", hipGetErrorString(_e), " at ", __FILE__, \
+                  ":", __LINE__);                                                \
+    }                                                                            \
+  } while (0)
+
+static inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
+
+// splitmix64 — counter-based RNG hash for dropout (deterministic per
+// (seed, offset, index); quality is ample for Bernoulli masks).
+__device__ __forceinline__ uint64_t slk_mix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+__device__ __forceinline__ float slk_uniform(uint64_t seed, uint64_t offset,
+                                             uint64_t idx) {
+  uint64_t h = slk_mix64(seed ^ slk_mix64(offset * 0xD1342543DE82EF95ull + idx));
+  // top 24 bits -> [0, 1)
+  return (float)(h >> 40) * (1.0f / 16777216.0f);
+}
+
+// block-level reduction into lane 0 of wave 0 (sums `val` over blockDim.x
+// threads; blockDim.x must be a multiple of 64 and <= 1024)
+template <typename T>
+__device__ __forceinline__ T slk_block_sum(T val, T* lds_scratch /* >= 16 */) {
+  // wave reduce
+  for (int off = 32; off > 0; off >>= 1) val += __shfl_down(val, off, 64);
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int nwaves = blockDim.x >> 6;
+  if (lane == 0) lds_scratch[wid] = val;
+  __syncthreads();
+  T out = (T)0;
+  if (wid == 0) {
+    out = (lane < nwaves) ? lds_scratch[lane] : (T)0;
+    for (int off = 8; off > 0; off >>= 1) out += __shfl_down(out, off, 64);
+  }
+  return out;  // valid in thread 0 only
+}

@@ -1,0 +1,134 @@
+// Strided/batched fp32 MFMA GEMM + Linear forward + column-sum.
+// Serves Linear fwd/bwd (reference nn.Linear sites: src/model/VGG16_CIFAR10.py:107-117),
+// attention QK^T / PV batched matmuls (src/model/BERT_AGNEWS.py:66-74), and
+// LoRA adapters.  Transposes are handled by stride swaps — no data movement.
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+
+#include "tile_gemm.h"
+
+namespace slk {
+
+struct StridedGather {
+  const float* A;
+  const float* B;
+  long sAb, sAm, sAk;
+  long sBb, sBk, sBn;
+  __device__ float loadA(int b, int m, int k) const {
+    return A[(long)b * sAb + (long)m * sAm + (long)k * sAk];
+  }
+  __device__ float loadB(int b, int k, int n) const {
+    return B[(long)b * sBb + (long)k * sBk + (long)n * sBn];
+  }
+};
+
+struct StridedStore {
+  float* C;
+  long sCb;
+  int N;
+  const float* bias;  // per-column bias (nullable)
+  bool accumulate;    // atomic accumulate (split-K or beta=1)
+  __device__ void store(int b, int m, int n, float v) const {
+    float* p = C + (long)b * sCb + (long)m * N + n;
+    if (bias != nullptr) v += bias[n];
+    if (accumulate) {
+      atomicAdd(p, v);
+    } else {
+      *p = v;
+    }
+  }
+};
+
+static void launch_strided(const at::Tensor& a2, const at::Tensor& b2, at::Tensor& c,
+                           bool ta, bool tb, const c10::optional<at::Tensor>& bias,
+                           bool accumulate, int split_k) {
+  const bool batched = a2.dim() == 3;
+  const int nb = batched ? a2.size(0) : 1;
+  const int M = ta ? a2.size(-1) : a2.size(-2);
+  const int K = ta ? a2.size(-2) : a2.size(-1);
+  const int N = tb ? b2.size(-2) : b2.size(-1);
+
+  StridedGather g;
+  g.A = a2.data_ptr<float>();
+  g.B = b2.data_ptr<float>();
+  g.sAb = batched ? a2.stride(0) : 0;
+  g.sAm = ta ? a2.stride(-1) : a2.stride(-2);
+  g.sAk = ta ? a2.stride(-2) : a2.stride(-1);
+  g.sBb = (b2.dim() == 3) ? b2.stride(0) : 0;
+  g.sBk = tb ? b2.stride(-1) : b2.stride(-2);
+  g.sBn = tb ? b2.stride(-2) : b2.stride(-1);
+
+  StridedStore st;
+  st.C = c.data_ptr<float>();
+  st.sCb = batched ? (long)M * N : 0;
+  st.N = N;
+  st.bias = bias.has_value() ? bias->data_ptr<float>() : nullptr;
+  st.accumulate = accumulate || split_k > 1;
+
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  slk_launch_gemm(g, st, M, N, K, nb, split_k, stream);
+}
+
+at::Tensor matmul_f32(const at::Tensor& a, const at::Tensor& b, bool ta, bool tb,
+                      c10::optional<at::Tensor> out, bool accumulate) {
+  TORCH_CHECK(a.is_cuda() && b.is_cuda(), "matmul_f32: GPU tensors required");
+  TORCH_CHECK(a.scalar_type() == at::kFloat && b.scalar_type() == at::kFloat,
+              "matmul_f32 is fp32-only");
+  TORCH_CHECK(a.dim() == b.dim() && (a.dim() == 2 || a.dim() == 3),
+              "matmul_f32: 2-D or 3-D tensors, same rank");
+  const int M = ta ? a.size(-1) : a.size(-2);
+  const int Ka = ta ? a.size(-2) : a.size(-1);
+  const int Kb = tb ? b.size(-1) : b.size(-2);
+  const int N = tb ? b.size(-2) : b.size(-1);
+  TORCH_CHECK(Ka == Kb, "matmul_f32: inner dims mismatch ", Ka, " vs ", Kb);
+  if (a.dim() == 3) TORCH_CHECK(a.size(0) == b.size(0), "batch mismatch");
+
+  at::Tensor c;
+  if (out.has_value()) {
+    c = *out;
+  } else {
+    c = a.dim() == 3 ? at::empty({a.size(0), M, N}, a.options())
+                     : at::empty({M, N}, a.options());
+  }
+  int split_k = 1;
+  if (!accumulate && !out.has_value()) {
+    split_k = slk_pick_split_k(M, N, Ka, a.dim() == 3 ? a.size(0) : 1);
+    if (split_k > 1) c.zero_();
+  }
+  launch_strided(a, b, c, ta, tb, c10::nullopt, accumulate, split_k);
+  return c;
+}
+
+at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
+                      c10::optional<at::Tensor> bias) {
+  // y[M,N] = x[M,K] @ w[N,K]^T + b
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat, "linear_fwd: fp32 GPU");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1));
+  auto y = at::empty({x.size(0), w.size(0)}, x.options());
+  launch_strided(x, w, y, /*ta=*/false, /*tb=*/true, bias, /*acc=*/false, 1);
+  return y;
+}
+
+// column sum of a [M,N] matrix -> [N] (Linear bias gradient)
+__global__ void colsum_kernel(const float* __restrict__ x, float* __restrict__ out,
+                              int M, int N) {
+  int n = blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= N) return;
+  float acc = 0.f;
+  for (int m = 0; m < M; ++m) acc += x[(long)m * N + n];
+  out[n] = acc;
+}
+
+at::Tensor colsum_f32(const at::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.scalar_type() == at::kFloat);
+  auto xc = x.contiguous();
+  auto out = at::empty({x.size(1)}, x.options());
+  int N = x.size(1);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(colsum_kernel, dim3(ceil_div(N, 256)), dim3(256), 0, stream,
+                     xc.data_ptr<float>(), out.data_ptr<float>(), x.size(0), N);
+  return out;
+}
+
+}  // namespace slk

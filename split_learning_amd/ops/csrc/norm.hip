@@ -1,0 +1,296 @@
+// BatchNorm2d (train/eval, fwd/bwd) and LayerNorm (fwd/bwd) for fp32 NCHW /
+// row-major tensors.  BatchNorm keeps exact torch semantics (biased batch var
+// for normalisation, unbiased for running stats — done host-side in
+// ops/functional.py) so state_dict running stats stay .pth-compatible
+// (SURVEY.md §7 hard-part 2).
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.h"
+
+namespace slk {
+
+// ---------------- BatchNorm2d ----------------
+
+// one block per channel; double accumulators for stable mean/var at any N*H*W
+__global__ void bn_stats_kernel(const float* __restrict__ x, float* __restrict__ mean,
+                                float* __restrict__ var, int B, int C, int HW) {
+  __shared__ double scratch[16];
+  const int c = blockIdx.x;
+  double s = 0.0, s2 = 0.0;
+  for (int i = threadIdx.x; i < B * HW; i += blockDim.x) {
+    const int b = i / HW;
+    const int r = i - b * HW;
+    const double v = (double)x[((long)b * C + c) * HW + r];
+    s += v;
+    s2 += v * v;
+  }
+  double ts = slk_block_sum(s, scratch);
+  __syncthreads();
+  double ts2 = slk_block_sum(s2, scratch);
+  if (threadIdx.x == 0) {
+    const double n = (double)B * HW;
+    const double m = ts / n;
+    mean[c] = (float)m;
+    var[c] = (float)fmax(ts2 / n - m * m, 0.0);
+  }
+}
+
+__global__ void bn_fwd_kernel(const float* __restrict__ x, float* __restrict__ y,
+                              const float* __restrict__ mean,
+                              const float* __restrict__ invstd,
+                              const float* __restrict__ gamma,
+                              const float* __restrict__ beta, long total, int C,
+                              int HW, bool relu) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const int c = (int)((i / HW) % C);
+    float v = (x[i] - mean[c]) * invstd[c] * gamma[c] + beta[c];
+    if (relu) v = fmaxf(v, 0.f);
+    y[i] = v;
+  }
+}
+
+// reductions for backward: sum(gy) and sum(gy * xhat) per channel
+__global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
+                                     const float* __restrict__ gy,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     float* __restrict__ sum_gy,
+                                     float* __restrict__ sum_gy_xhat, int B, int C,
+                                     int HW) {
+  __shared__ double scratch[16];
+  const int c = blockIdx.x;
+  const float m = mean[c], is = invstd[c];
+  double s = 0.0, sx = 0.0;
+  for (int i = threadIdx.x; i < B * HW; i += blockDim.x) {
+    const int b = i / HW;
+    const int r = i - b * HW;
+    const long off = ((long)b * C + c) * HW + r;
+    const double g = (double)gy[off];
+    s += g;
+    sx += g * (double)((x[off] - m) * is);
+  }
+  double ts = slk_block_sum(s, scratch);
+  __syncthreads();
+  double tsx = slk_block_sum(sx, scratch);
+  if (threadIdx.x == 0) {
+    sum_gy[c] = (float)ts;
+    sum_gy_xhat[c] = (float)tsx;
+  }
+}
+
+__global__ void bn_bwd_dx_kernel(const float* __restrict__ x,
+                                 const float* __restrict__ gy,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ invstd,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ sum_gy,
+                                 const float* __restrict__ sum_gy_xhat,
+                                 float* __restrict__ gx, long total, int C, int HW,
+                                 float inv_n, bool training) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const int c = (int)((i / HW) % C);
+    const float is = invstd[c];
+    if (training) {
+      const float xhat = (x[i] - mean[c]) * is;
+      gx[i] = gamma[c] * is *
+              (gy[i] - sum_gy[c] * inv_n - xhat * sum_gy_xhat[c] * inv_n);
+    } else {
+      gx[i] = gy[i] * gamma[c] * is;
+    }
+  }
+}
+
+std::vector<at::Tensor> bn2d_stats(const at::Tensor& x) {
+  const int B = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  auto mean = at::empty({C}, x.options());
+  auto var = at::empty({C}, x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(bn_stats_kernel, dim3(C), dim3(256), 0, stream,
+                     x.data_ptr<float>(), mean.data_ptr<float>(),
+                     var.data_ptr<float>(), B, C, HW);
+  return {mean, var};
+}
+
+at::Tensor bn2d_fwd(const at::Tensor& x, const at::Tensor& mean,
+                    const at::Tensor& invstd, const at::Tensor& gamma,
+                    const at::Tensor& beta, bool relu) {
+  auto y = at::empty_like(x);
+  const long total = x.numel();
+  const int C = x.size(1), HW = x.size(2) * x.size(3);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int grid = (int)std::min<long>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(bn_fwd_kernel, dim3(grid), dim3(256), 0, stream,
+                     x.data_ptr<float>(), y.data_ptr<float>(),
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(), total, C, HW,
+                     relu);
+  return y;
+}
+
+static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tensor& gy,
+                                             const at::Tensor& gamma,
+                                             const at::Tensor& mean,
+                                             const at::Tensor& invstd, bool training) {
+  const int B = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  const long total = x.numel();
+  auto sum_gy = at::empty({C}, x.options());
+  auto sum_gy_xhat = at::empty({C}, x.options());
+  auto gx = at::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C), dim3(256), 0, stream,
+                     x.data_ptr<float>(), gy.data_ptr<float>(),
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     sum_gy.data_ptr<float>(), sum_gy_xhat.data_ptr<float>(), B, C,
+                     HW);
+  int grid = (int)std::min<long>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(grid), dim3(256), 0, stream,
+                     x.data_ptr<float>(), gy.data_ptr<float>(),
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     gamma.data_ptr<float>(), sum_gy.data_ptr<float>(),
+                     sum_gy_xhat.data_ptr<float>(), gx.data_ptr<float>(), total, C,
+                     HW, 1.0f / (float)((long)B * HW), training);
+  // ggamma = sum_gy_xhat, gbeta = sum_gy
+  return {gx, sum_gy_xhat, sum_gy};
+}
+
+std::vector<at::Tensor> bn2d_bwd(const at::Tensor& x, const at::Tensor& gy,
+                                 const at::Tensor& gamma, const at::Tensor& mean,
+                                 const at::Tensor& invstd) {
+  return bn2d_bwd_impl(x, gy, gamma, mean, invstd, true);
+}
+
+std::vector<at::Tensor> bn2d_bwd_eval(const at::Tensor& x, const at::Tensor& gy,
+                                      const at::Tensor& gamma, const at::Tensor& mean,
+                                      const at::Tensor& invstd) {
+  return bn2d_bwd_impl(x, gy, gamma, mean, invstd, false);
+}
+
+// ---------------- LayerNorm (last-dim) ----------------
+
+// one block per row (rows up to ~4K cols; looping supports any D)
+__global__ void ln_fwd_kernel(const float* __restrict__ x, float* __restrict__ y,
+                              const float* __restrict__ gamma,
+                              const float* __restrict__ beta,
+                              float* __restrict__ mean_out,
+                              float* __restrict__ invstd_out, int D, float eps) {
+  __shared__ float scratch[16];
+  __shared__ float s_mean, s_invstd;
+  const long row = blockIdx.x;
+  const float* xr = x + row * D;
+  float s = 0.f;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) s += xr[d];
+  float total = slk_block_sum(s, scratch);
+  if (threadIdx.x == 0) s_mean = total / D;
+  __syncthreads();
+  const float m = s_mean;
+  float v = 0.f;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    const float t = xr[d] - m;
+    v += t * t;
+  }
+  float vtotal = slk_block_sum(v, scratch);
+  if (threadIdx.x == 0) s_invstd = rsqrtf(vtotal / D + eps);
+  __syncthreads();
+  const float is = s_invstd;
+  float* yr = y + row * D;
+  for (int d = threadIdx.x; d < D; d += blockDim.x)
+    yr[d] = (xr[d] - m) * is * gamma[d] + beta[d];
+  if (threadIdx.x == 0) {
+    mean_out[row] = m;
+    invstd_out[row] = is;
+  }
+}
+
+__global__ void ln_bwd_dx_kernel(const float* __restrict__ gy,
+                                 const float* __restrict__ x,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ invstd,
+                                 float* __restrict__ gx, int D) {
+  __shared__ float scratch[16];
+  __shared__ float s_a, s_b;
+  const long row = blockIdx.x;
+  const float* xr = x + row * D;
+  const float* gr = gy + row * D;
+  const float m = mean[row], is = invstd[row];
+  float a = 0.f, b = 0.f;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    const float gg = gr[d] * gamma[d];
+    a += gg;
+    b += gg * (xr[d] - m) * is;
+  }
+  float ta = slk_block_sum(a, scratch);
+  if (threadIdx.x == 0) s_a = ta / D;
+  __syncthreads();  // scratch reuse barrier between the two block sums
+  float tb = slk_block_sum(b, scratch);
+  if (threadIdx.x == 0) s_b = tb / D;
+  __syncthreads();
+  float* oxr = gx + row * D;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    const float xhat = (xr[d] - m) * is;
+    oxr[d] = is * (gr[d] * gamma[d] - s_a - xhat * s_b);
+  }
+}
+
+// column reductions for dgamma/dbeta: thread per column, loop rows (coalesced)
+__global__ void ln_bwd_dgamma_kernel(const float* __restrict__ gy,
+                                     const float* __restrict__ x,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     float* __restrict__ dgamma,
+                                     float* __restrict__ dbeta, long R, int D) {
+  const int d = blockIdx.x * blockDim.x + threadIdx.x;
+  if (d >= D) return;
+  float sg = 0.f, sb = 0.f;
+  for (long r = 0; r < R; ++r) {
+    const float g = gy[r * D + d];
+    sg += g * (x[r * D + d] - mean[r]) * invstd[r];
+    sb += g;
+  }
+  dgamma[d] = sg;
+  dbeta[d] = sb;
+}
+
+std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x, const at::Tensor& gamma,
+                                      const at::Tensor& beta, double eps) {
+  const long R = x.size(0);
+  const int D = x.size(1);
+  auto y = at::empty_like(x);
+  auto mean = at::empty({R}, x.options());
+  auto invstd = at::empty({R}, x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int threads = D >= 256 ? 256 : 64;
+  hipLaunchKernelGGL(ln_fwd_kernel, dim3(R), dim3(threads), 0, stream,
+                     x.data_ptr<float>(), y.data_ptr<float>(),
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(), D,
+                     (float)eps);
+  return {y, mean, invstd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(const at::Tensor& gy, const at::Tensor& x,
+                                      const at::Tensor& gamma, const at::Tensor& mean,
+                                      const at::Tensor& invstd) {
+  const long R = x.size(0);
+  const int D = x.size(1);
+  auto gx = at::empty_like(x);
+  auto dgamma = at::empty({D}, x.options());
+  auto dbeta = at::empty({D}, x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int threads = D >= 256 ? 256 : 64;
+  hipLaunchKernelGGL(ln_bwd_dx_kernel, dim3(R), dim3(threads), 0, stream,
+                     gy.data_ptr<float>(), x.data_ptr<float>(),
+                     gamma.data_ptr<float>(), mean.data_ptr<float>(),
+                     invstd.data_ptr<float>(), gx.data_ptr<float>(), D);
+  hipLaunchKernelGGL(ln_bwd_dgamma_kernel, dim3(ceil_div(D, 256)), dim3(256), 0,
+                     stream, gy.data_ptr<float>(), x.data_ptr<float>(),
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), R, D);
+  return {gx, dgamma, dbeta};
+}
+
+}  // namespace slk

@@ -1,0 +1,129 @@
+// MFMA f32 tile-GEMM framework for gfx950 (CDNA4).
+//
+// One templated kernel computes C = A @ B for fp32 operands on the exact-f32
+// MFMA path (v_mfma_f32_16x16x4_f32 — bitwise an fmaf chain, §3 of the CDNA
+// guide), with pluggable gather functors so the same MFMA/LDS core serves:
+//   * plain strided (batched) GEMM               (gemm_f32.hip)
+//   * implicit-GEMM conv2d fwd / bwd-data / bwd-weight (conv2d.hip)
+//
+// Geometry: 256 threads = 4 waves; block tile 64(M) x 64(N); BK=16 K-step;
+// each wave owns a 32x32 sub-tile = 2x2 MFMA fragments with 4 f32x4
+// accumulators.  A and B tiles are staged k-major in LDS ([BK][64+pad]) so the
+// MFMA operand reads (lane&15 consecutive) are conflict-light.  Split-K over
+// grid.z with atomic C accumulation is supported for K-heavy shapes
+// (conv bwd-weight: K = B*OH*OW).
+#pragma once
+
+#include "common.h"
+
+constexpr int SLK_BM = 64;
+constexpr int SLK_BN = 64;
+constexpr int SLK_BK = 16;
+constexpr int SLK_LDS_PAD = 4;
+
+// Each gather functor provides:
+//   __device__ float loadA(int batch, int m, int k) const;  // 0 if OOB
+//   __device__ float loadB(int batch, int k, int n) const;
+// The store functor provides:
+//   __device__ void store(int batch, int m, int n, float v) const; // handles OOB
+template <typename Gather, typename Store>
+__global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
+    Gather g, Store st, int M, int N, int K, int split_k, int k_per_split) {
+  __shared__ float ldsA[SLK_BK][SLK_BM + SLK_LDS_PAD];
+  __shared__ float ldsB[SLK_BK][SLK_BN + SLK_LDS_PAD];
+
+  const int tile_n = blockIdx.x;
+  const int tile_m = blockIdx.y;
+  const int batch = blockIdx.z / split_k;
+  const int ks = blockIdx.z % split_k;
+
+  const int m0 = tile_m * SLK_BM;
+  const int n0 = tile_n * SLK_BN;
+  const int k_begin = ks * k_per_split;
+  const int k_end = min(K, k_begin + k_per_split);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * 32;  // wave row offset in tile
+  const int wn = (wid & 1) * 32;   // wave col offset in tile
+
+  const int frag_r = lane >> 4;    // 0..3: k sub-index for A/B operands
+  const int frag_c = lane & 15;    // 0..15
+
+  f32x4 acc[2][2] = {};
+
+  for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
+    // stage A[k][m] (k-fast gather for contiguous global rows)
+    #pragma unroll
+    for (int i = 0; i < (SLK_BM * SLK_BK) / 256; ++i) {
+      int idx = tid + i * 256;
+      int m = idx >> 4;          // 0..63
+      int k = idx & 15;          // 0..15
+      ldsA[k][m] = (k0 + k < k_end && m0 + m < M) ? g.loadA(batch, m0 + m, k0 + k)
+                                                  : 0.0f;
+    }
+    // stage B[k][n] (n-fast: coalesced for row-major B)
+    #pragma unroll
+    for (int i = 0; i < (SLK_BN * SLK_BK) / 256; ++i) {
+      int idx = tid + i * 256;
+      int k = idx >> 6;          // 0..15
+      int n = idx & 63;          // 0..63
+      ldsB[k][n] = (k0 + k < k_end && n0 + n < N) ? g.loadB(batch, k0 + k, n0 + n)
+                                                  : 0.0f;
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int kk = 0; kk < SLK_BK / 4; ++kk) {
+      const int kr = kk * 4 + frag_r;
+      float a0 = ldsA[kr][wm + frag_c];
+      float a1 = ldsA[kr][wm + 16 + frag_c];
+      float b0 = ldsB[kr][wn + frag_c];
+      float b1 = ldsB[kr][wn + 16 + frag_c];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C/D fragment mapping for 16x16x4: col = lane&15, row = (lane>>4)*4 + i
+  #pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+    #pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int row = m0 + wm + mi * 16 + frag_r * 4 + i;
+        int col = n0 + wn + ni * 16 + frag_c;
+        if (row < M && col < N) st.store(batch, row, col, acc[mi][ni][i]);
+      }
+    }
+  }
+}
+
+// Host-side launch helper.
+template <typename Gather, typename Store>
+inline void slk_launch_gemm(const Gather& g, const Store& st, int M, int N, int K,
+                            int n_batch, int split_k, hipStream_t stream) {
+  if (split_k < 1) split_k = 1;
+  int k_per_split = ceil_div(K, split_k);
+  // round k_per_split up to a BK multiple so every split starts aligned
+  k_per_split = ((k_per_split + SLK_BK - 1) / SLK_BK) * SLK_BK;
+  split_k = ceil_div(K, k_per_split);
+  dim3 grid(ceil_div(N, SLK_BN), ceil_div(M, SLK_BM), n_batch * split_k);
+  hipLaunchKernelGGL((slk_mfma_gemm_kernel<Gather, Store>), grid, dim3(256), 0,
+                     stream, g, st, M, N, K, split_k, k_per_split);
+}
+
+// Heuristic: pick split_k so the grid roughly fills 256 CUs (8 XCDs).
+inline int slk_pick_split_k(int M, int N, int K, int n_batch) {
+  long tiles = (long)ceil_div(M, SLK_BM) * ceil_div(N, SLK_BN) * (n_batch > 0 ? n_batch : 1);
+  if (tiles >= 256 || K <= SLK_BK * 2) return 1;
+  long want = 256 / tiles;
+  long maxk = (K + 4 * SLK_BK - 1) / (4 * SLK_BK);  // keep >=4 BK steps per split
+  long sk = want < maxk ? want : maxk;
+  return (int)(sk < 1 ? 1 : sk);
+}

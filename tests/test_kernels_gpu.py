@@ -1,0 +1,336 @@
+"""Numerics tests: every HIP kernel vs the plain PyTorch fp32 reference.
+
+All tests are @pytest.mark.gpu (run on a real MI355X via gpurun / at round end).
+Tolerances: the MFMA f32 path is an exact fmaf chain, but summation order
+differs from torch's, so we compare with atol/rtol scaled to operand magnitude.
+"""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+
+def _native():
+    from split_learning_amd.ops import native
+    return native()
+
+
+def assert_close(a, b, atol=1e-4, rtol=1e-4, what=""):
+    torch.testing.assert_close(a, b, atol=atol, rtol=rtol, msg=lambda m: f"{what}: {m}")
+
+
+# ---------------- GEMM ----------------
+
+@pytest.mark.parametrize("m,n,k", [(32, 64, 27), (32, 4096, 512), (100, 10, 4096),
+                                   (128, 128, 128), (33, 65, 17), (64, 64, 4608)])
+def test_matmul_plain(m, n, k):
+    ext = _native()
+    a = torch.randn(m, k, device="cuda")
+    b = torch.randn(k, n, device="cuda")
+    c = ext.matmul_f32(a, b, False, False, None, False)
+    assert_close(c, a @ b, atol=1e-3, rtol=1e-3, what=f"matmul {m}x{n}x{k}")
+
+
+@pytest.mark.parametrize("ta,tb", [(False, False), (True, False), (False, True), (True, True)])
+def test_matmul_trans(ta, tb):
+    ext = _native()
+    m, n, k = 48, 80, 96
+    a = torch.randn((k, m) if ta else (m, k), device="cuda")
+    b = torch.randn((n, k) if tb else (k, n), device="cuda")
+    ref = (a.t() if ta else a) @ (b.t() if tb else b)
+    c = ext.matmul_f32(a, b, ta, tb, None, False)
+    assert_close(c, ref, atol=1e-3, rtol=1e-3, what=f"matmul ta={ta} tb={tb}")
+
+
+def test_matmul_batched():
+    ext = _native()
+    a = torch.randn(384, 128, 64, device="cuda")  # B*H, S, hd
+    b = torch.randn(384, 128, 64, device="cuda")
+    c = ext.matmul_f32(a, b, False, True, None, False)  # QK^T
+    assert_close(c, a @ b.transpose(-1, -2), atol=1e-3, rtol=1e-3, what="batched QK^T")
+
+
+def test_linear_fwd_bias():
+    ext = _native()
+    x = torch.randn(32, 512, device="cuda")
+    w = torch.randn(4096, 512, device="cuda")
+    b = torch.randn(4096, device="cuda")
+    assert_close(ext.linear_fwd(x, w, b), x @ w.t() + b, atol=1e-3, rtol=1e-3,
+                 what="linear")
+
+
+def test_colsum():
+    ext = _native()
+    x = torch.randn(517, 321, device="cuda")
+    assert_close(ext.colsum_f32(x), x.sum(0), atol=1e-3, rtol=1e-3, what="colsum")
+
+
+# ---------------- Conv2d (every VGG16 shape family + strided + 1x1 + patch) ----
+
+CONV_CASES = [
+    # (B, Ci, H, W, Co, K, stride, pad)
+    (4, 3, 32, 32, 64, 3, 1, 1),
+    (4, 64, 32, 32, 64, 3, 1, 1),
+    (4, 64, 16, 16, 128, 3, 1, 1),
+    (4, 256, 4, 4, 512, 3, 1, 1),
+    (4, 512, 2, 2, 512, 3, 1, 1),
+    (4, 64, 16, 16, 128, 1, 1, 0),    # MobileNet 1x1
+    (4, 64, 16, 16, 64, 3, 2, 1),     # MobileNet stride-2
+    (4, 3, 32, 32, 128, 4, 4, 0),     # ViT patch embed
+]
+
+
+@pytest.mark.parametrize("B,Ci,H,W,Co,K,s,p", CONV_CASES)
+def test_conv2d_fwd(B, Ci, H, W, Co, K, s, p):
+    ext = _native()
+    x = torch.randn(B, Ci, H, W, device="cuda")
+    w = torch.randn(Co, Ci, K, K, device="cuda")
+    b = torch.randn(Co, device="cuda")
+    y = ext.conv2d_fwd(x, w, b, s, p)
+    ref = F.conv2d(x, w, b, stride=s, padding=p)
+    assert_close(y, ref, atol=2e-3, rtol=2e-3, what=f"conv fwd {Ci}->{Co} k{K}s{s}")
+
+
+@pytest.mark.parametrize("B,Ci,H,W,Co,K,s,p", CONV_CASES)
+def test_conv2d_bwd(B, Ci, H, W, Co, K, s, p):
+    ext = _native()
+    x = torch.randn(B, Ci, H, W, device="cuda", requires_grad=True)
+    w = torch.randn(Co, Ci, K, K, device="cuda", requires_grad=True)
+    b = torch.randn(Co, device="cuda", requires_grad=True)
+    y = F.conv2d(x, w, b, stride=s, padding=p)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+
+    gx = ext.conv2d_bwd_data(gy, w.detach(), s, p, H, W)
+    gw = ext.conv2d_bwd_weight(gy, x.detach(), K, K, s, p)
+    gb = ext.conv2d_bwd_bias(gy)
+    assert_close(gx, x.grad, atol=2e-3, rtol=2e-3, what="conv bwd_data")
+    assert_close(gw, w.grad, atol=2e-2, rtol=2e-3, what="conv bwd_weight")
+    assert_close(gb, b.grad, atol=2e-3, rtol=2e-3, what="conv bwd_bias")
+
+
+# ---------------- BatchNorm ----------------
+
+def test_bn2d_train_fwd_bwd():
+    from split_learning_amd.ops import functional as hf
+    torch.manual_seed(0)
+    x = torch.randn(8, 32, 16, 16, device="cuda")
+    gamma = torch.randn(32, device="cuda", requires_grad=True)
+    beta = torch.randn(32, device="cuda", requires_grad=True)
+    rm = torch.zeros(32, device="cuda")
+    rv = torch.ones(32, device="cuda")
+
+    x1 = x.clone().requires_grad_(True)
+    y = hf.batch_norm2d(x1, gamma, beta, rm, rv, training=True)
+    # torch reference
+    x2 = x.clone().requires_grad_(True)
+    g2 = gamma.detach().clone().requires_grad_(True)
+    b2 = beta.detach().clone().requires_grad_(True)
+    rm2 = torch.zeros(32, device="cuda")
+    rv2 = torch.ones(32, device="cuda")
+    yref = F.batch_norm(x2, rm2, rv2, g2, b2, training=True, momentum=0.1, eps=1e-5)
+    assert_close(y, yref, atol=1e-4, rtol=1e-4, what="bn fwd")
+    assert_close(rm, rm2, atol=1e-4, rtol=1e-4, what="bn running_mean")
+    assert_close(rv, rv2, atol=1e-4, rtol=1e-4, what="bn running_var")
+
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    yref.backward(gy)
+    assert_close(x1.grad, x2.grad, atol=1e-4, rtol=1e-4, what="bn gx")
+    assert_close(gamma.grad, g2.grad, atol=1e-3, rtol=1e-3, what="bn ggamma")
+    assert_close(beta.grad, b2.grad, atol=1e-3, rtol=1e-3, what="bn gbeta")
+
+
+def test_bn2d_eval():
+    from split_learning_amd.ops import functional as hf
+    x = torch.randn(4, 16, 8, 8, device="cuda")
+    gamma = torch.randn(16, device="cuda")
+    beta = torch.randn(16, device="cuda")
+    rm = torch.randn(16, device="cuda")
+    rv = torch.rand(16, device="cuda") + 0.5
+    y = hf.batch_norm2d(x, gamma, beta, rm, rv, training=False)
+    yref = F.batch_norm(x, rm, rv, gamma, beta, training=False, eps=1e-5)
+    assert_close(y, yref, atol=1e-4, rtol=1e-4, what="bn eval")
+
+
+# ---------------- LayerNorm ----------------
+
+@pytest.mark.parametrize("R,D", [(256, 768), (99, 64), (32, 3072)])
+def test_layernorm(R, D):
+    from split_learning_amd.ops import functional as hf
+    x = torch.randn(R, D, device="cuda", requires_grad=True)
+    g = torch.randn(D, device="cuda", requires_grad=True)
+    b = torch.randn(D, device="cuda", requires_grad=True)
+    y = hf.layer_norm(x, g, b, eps=1e-12)
+    x2 = x.detach().clone().requires_grad_(True)
+    g2 = g.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    yref = F.layer_norm(x2, (D,), g2, b2, eps=1e-12)
+    assert_close(y, yref, atol=1e-4, rtol=1e-4, what="ln fwd")
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    yref.backward(gy)
+    assert_close(x.grad, x2.grad, atol=1e-4, rtol=1e-4, what="ln gx")
+    assert_close(g.grad, g2.grad, atol=1e-3, rtol=1e-3, what="ln dgamma")
+    assert_close(b.grad, b2.grad, atol=1e-3, rtol=1e-3, what="ln dbeta")
+
+
+# ---------------- elementwise ----------------
+
+def test_relu_gelu_tanh():
+    from split_learning_amd.ops import functional as hf
+    for fn, ref in [(hf.relu, F.relu), (hf.gelu, F.gelu), (hf.tanh, torch.tanh)]:
+        x = torch.randn(1000, device="cuda", requires_grad=True)
+        x2 = x.detach().clone().requires_grad_(True)
+        y, yref = fn(x), ref(x2)
+        assert_close(y, yref, atol=1e-5, rtol=1e-5, what=str(fn))
+        gy = torch.randn_like(y)
+        y.backward(gy)
+        yref.backward(gy)
+        assert_close(x.grad, x2.grad, atol=1e-5, rtol=1e-5, what=f"{fn} bwd")
+
+
+def test_maxpool():
+    from split_learning_amd.ops import functional as hf
+    for H, W in [(32, 32), (7, 7)]:
+        x = torch.randn(4, 8, H, W, device="cuda", requires_grad=True)
+        x2 = x.detach().clone().requires_grad_(True)
+        y = hf.maxpool2x2(x)
+        yref = F.max_pool2d(x2, 2, 2)
+        assert_close(y, yref, what=f"maxpool {H}x{W}")
+        gy = torch.randn_like(y)
+        y.backward(gy)
+        yref.backward(gy)
+        assert_close(x.grad, x2.grad, what=f"maxpool bwd {H}x{W}")
+
+
+def test_dropout_stats_and_bwd():
+    from split_learning_amd.ops import functional as hf
+    hf.seed_dropout(1234)
+    x = torch.ones(1 << 20, device="cuda", requires_grad=True)
+    y = hf.dropout(x, 0.5, training=True)
+    keep = (y > 0).float().mean().item()
+    assert abs(keep - 0.5) < 0.01
+    assert torch.allclose(y[y > 0], torch.full_like(y[y > 0], 2.0))
+    y.sum().backward()
+    assert torch.allclose(x.grad[y > 0], torch.full_like(x.grad[y > 0], 2.0))
+    assert torch.all(x.grad[y == 0] == 0)
+
+
+def test_softmax():
+    from split_learning_amd.ops import functional as hf
+    x = torch.randn(384, 128, device="cuda", requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    y = hf.softmax_lastdim(x)
+    yref = F.softmax(x2, dim=-1)
+    assert_close(y, yref, atol=1e-5, rtol=1e-4, what="softmax")
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    yref.backward(gy)
+    assert_close(x.grad, x2.grad, atol=1e-5, rtol=1e-4, what="softmax bwd")
+
+
+def test_cross_entropy():
+    from split_learning_amd.ops import functional as hf
+    logits = torch.randn(32, 10, device="cuda", requires_grad=True)
+    labels = torch.randint(0, 10, (32,), device="cuda")
+    l2 = logits.detach().clone().requires_grad_(True)
+    loss = hf.cross_entropy(logits, labels)
+    lref = F.cross_entropy(l2, labels)
+    assert_close(loss, lref, atol=1e-5, rtol=1e-5, what="ce loss")
+    loss.backward()
+    lref.backward()
+    assert_close(logits.grad, l2.grad, atol=1e-5, rtol=1e-5, what="ce grad")
+
+
+def test_embedding():
+    from split_learning_amd.ops import functional as hf
+    w = torch.randn(1000, 64, device="cuda", requires_grad=True)
+    w2 = w.detach().clone().requires_grad_(True)
+    ids = torch.randint(0, 1000, (8, 32), device="cuda")
+    ids[0, 0] = 0  # padding idx
+    y = hf.embedding(ids, w, padding_idx=0)
+    yref = F.embedding(ids, w2, padding_idx=0)
+    assert_close(y, yref, what="embedding")
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    yref.backward(gy)
+    assert_close(w.grad, w2.grad, atol=1e-4, rtol=1e-4, what="embedding bwd")
+
+
+# ---------------- optimizers ----------------
+
+def test_sgd_matches_torch():
+    from split_learning_amd.ops import functional as hf
+    torch.manual_seed(0)
+    p1 = [torch.randn(100, device="cuda") for _ in range(3)]
+    p2 = [p.clone() for p in p1]
+    g = [torch.randn(100, device="cuda") for _ in range(3)]
+    bufs = [torch.zeros_like(p) for p in p1]
+
+    tp = [p.clone().requires_grad_(True) for p in p2]
+    opt = torch.optim.SGD(tp, lr=0.01, momentum=0.5)
+    for step in range(3):
+        for t, gr in zip(tp, g):
+            t.grad = gr.clone()
+        opt.step()
+        hf.sgd_step(p1, g, bufs, lr=0.01, momentum=0.5, first_step=(step == 0))
+    for a, b in zip(p1, tp):
+        assert_close(a, b.detach(), atol=1e-6, rtol=1e-6, what="sgd")
+
+
+def test_adamw_matches_torch():
+    from split_learning_amd.ops import functional as hf
+    torch.manual_seed(0)
+    p1 = [torch.randn(100, device="cuda") for _ in range(2)]
+    g = [torch.randn(100, device="cuda") for _ in range(2)]
+    m = [torch.zeros_like(p) for p in p1]
+    v = [torch.zeros_like(p) for p in p1]
+    tp = [p.clone().requires_grad_(True) for p in p1]
+    opt = torch.optim.AdamW(tp, lr=5e-4, weight_decay=0.01)
+    for step in range(1, 4):
+        for t, gr in zip(tp, g):
+            t.grad = gr.clone()
+        opt.step()
+        hf.adamw_step(p1, g, m, v, step=step, lr=5e-4, beta1=0.9, beta2=0.999,
+                      eps=1e-8, weight_decay=0.01)
+    for a, b in zip(p1, tp):
+        assert_close(a, b.detach(), atol=1e-5, rtol=1e-5, what="adamw")
+
+
+# ---------------- end-to-end model parity on GPU ----------------
+
+def test_vgg16_forward_matches_cpu():
+    """GPU (native kernels) forward == CPU (torch) forward in eval mode."""
+    from split_learning_amd.models import build_partition
+    torch.manual_seed(0)
+    model = build_partition("VGG16", "CIFAR10", [0, 0]).eval()
+    x = torch.randn(4, 3, 32, 32)
+    with torch.no_grad():
+        y_cpu = model(x)
+        y_gpu = model.cuda()(x.cuda())
+    assert_close(y_gpu.cpu(), y_cpu, atol=5e-3, rtol=5e-3, what="vgg16 cpu-vs-gpu")
+
+
+def test_bert_forward_matches_cpu():
+    from split_learning_amd.models import build_partition
+    torch.manual_seed(0)
+    model = build_partition("BERT", "AGNEWS", [0, 0]).eval()
+    ids = torch.randint(0, 28996, (2, 128))
+    with torch.no_grad():
+        y_cpu = model(ids)
+        y_gpu = model.cuda()(ids.cuda())
+    assert_close(y_gpu.cpu(), y_cpu, atol=5e-3, rtol=5e-3, what="bert cpu-vs-gpu")
+
+
+def test_kwt_forward_matches_cpu():
+    from split_learning_amd.models import build_partition
+    torch.manual_seed(0)
+    model = build_partition("KWT", "SPEECHCOMMANDS", [0, 0]).eval()
+    x = torch.randn(2, 40, 98)
+    with torch.no_grad():
+        y_cpu = model(x)
+        y_gpu = model.cuda()(x.cuda())
+    assert_close(y_gpu.cpu(), y_cpu, atol=5e-3, rtol=5e-3, what="kwt cpu-vs-gpu")
