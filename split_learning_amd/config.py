@@ -85,6 +85,10 @@ DEFAULT_CONFIG: Dict[str, Any] = {
     # "main" | "vanilla" | "cluster_fsl" | "dcsl" | "flex" | "2ls"
     "scheduler": {
         "policy": "main",
+        # stage-1 backward recomputes the forward with current weights
+        # (reference semantics, src/train/VGG16.py:89-91); False stashes the
+        # autograd graph (valid only with control-count 1)
+        "recompute": True,
         # Vanilla_SL / Cluster_FSL extras:
         "epochs": 1,
         "limited-time": None,         # wall-clock seconds cap per round (Vanilla)

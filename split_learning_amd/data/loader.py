@@ -1,0 +1,33 @@
+"""data_loader(data_name, batch_size, distribution, train) — same call surface
+as reference src/dataset/dataloader.py:124-134.
+
+Resolution order: real local dataset files (data/real.py) if present, else
+synthetic tensors of the same shape (data/synthetic.py).  `distribution` is the
+per-label sample-count list the server assigns to layer-1 clients.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+from torch.utils.data import DataLoader, TensorDataset
+
+from .synthetic import synthetic_tensors
+
+
+def data_loader(data_name: str, batch_size: int,
+                distribution: Optional[List[int]] = None, train: bool = True,
+                drop_last: bool = True, seed: int = 0) -> DataLoader:
+    try:
+        from .real import load_real
+        real = load_real(data_name, distribution, train)
+    except Exception:
+        real = None
+    if real is not None:
+        x, y = real
+    else:
+        x, y = synthetic_tensors(data_name, distribution, seed=seed)
+    ds = TensorDataset(x, y)
+    return DataLoader(ds, batch_size=batch_size, shuffle=train, drop_last=drop_last,
+                      num_workers=0)

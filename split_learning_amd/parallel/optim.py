@@ -1,0 +1,103 @@
+"""Optimizer wrappers: fused HIP multi-tensor kernels on GPU, torch on CPU.
+
+Semantics match the reference's optimizer choices exactly:
+SGD(lr, momentum) for VGG16/MobileNet/ViT (src/train/VGG16.py:62),
+AdamW(lr, weight_decay) for BERT/KWT (src/train/BERT.py:69, KWT.py:62).
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+import torch
+
+from ..ops import functional as hf
+
+
+class FusedSGD:
+    def __init__(self, params, lr: float, momentum: float = 0.0,
+                 weight_decay: float = 0.0):
+        self.params: List[torch.Tensor] = [p for p in params if p.requires_grad]
+        self.lr = lr
+        self.momentum = momentum
+        self.weight_decay = weight_decay
+        self.bufs = [torch.zeros_like(p) for p in self.params]
+        self.steps = 0
+        self._torch_opt = None
+
+    def zero_grad(self):
+        for p in self.params:
+            if p.grad is not None:
+                p.grad.zero_()
+
+    @torch.no_grad()
+    def step(self):
+        live = [(p, b) for p, b in zip(self.params, self.bufs) if p.grad is not None]
+        if not live:
+            return
+        if live[0][0].is_cuda:
+            hf.sgd_step([p for p, _ in live], [p.grad for p, _ in live],
+                        [b for _, b in live], self.lr, self.momentum,
+                        self.weight_decay, first_step=(self.steps == 0))
+        else:
+            for p, buf in live:
+                g = p.grad
+                if self.weight_decay:
+                    g = g + self.weight_decay * p
+                if self.momentum:
+                    if self.steps == 0:
+                        buf.copy_(g)
+                    else:
+                        buf.mul_(self.momentum).add_(g)
+                    g = buf
+                p.add_(g, alpha=-self.lr)
+        self.steps += 1
+
+
+class FusedAdamW:
+    def __init__(self, params, lr: float, weight_decay: float = 0.01,
+                 betas=(0.9, 0.999), eps: float = 1e-8):
+        self.params = [p for p in params if p.requires_grad]
+        self.lr = lr
+        self.weight_decay = weight_decay
+        self.beta1, self.beta2 = betas
+        self.eps = eps
+        self.m = [torch.zeros_like(p) for p in self.params]
+        self.v = [torch.zeros_like(p) for p in self.params]
+        self.steps = 0
+
+    def zero_grad(self):
+        for p in self.params:
+            if p.grad is not None:
+                p.grad.zero_()
+
+    @torch.no_grad()
+    def step(self):
+        self.steps += 1
+        live = [(p, m, v) for p, m, v in zip(self.params, self.m, self.v)
+                if p.grad is not None]
+        if not live:
+            return
+        if live[0][0].is_cuda:
+            hf.adamw_step([p for p, _, _ in live], [p.grad for p, _, _ in live],
+                          [m for _, m, _ in live], [v for _, _, v in live],
+                          self.steps, self.lr, self.beta1, self.beta2, self.eps,
+                          self.weight_decay)
+        else:
+            b1, b2 = self.beta1, self.beta2
+            bc1 = 1 - b1 ** self.steps
+            bc2 = 1 - b2 ** self.steps
+            for p, m, v in live:
+                p.mul_(1 - self.lr * self.weight_decay)
+                m.mul_(b1).add_(p.grad, alpha=1 - b1)
+                v.mul_(b2).addcmul_(p.grad, p.grad, value=1 - b2)
+                p.addcdiv_(m / bc1, (v / bc2).sqrt().add_(self.eps), value=-self.lr)
+
+
+def make_optimizer(model_name: str, params, learning: dict):
+    """Reference mapping: VGG16/MobileNetv1/ViT -> SGD+momentum; BERT/KWT -> AdamW."""
+    if model_name in ("BERT", "KWT"):
+        return FusedAdamW(params, lr=learning["learning-rate"],
+                          weight_decay=learning.get("weight-decay", 0.01))
+    return FusedSGD(params, lr=learning["learning-rate"],
+                    momentum=learning.get("momentum", 0.0))

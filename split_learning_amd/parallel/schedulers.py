@@ -1,0 +1,216 @@
+"""Stage training loops — the hot path of the engine.
+
+The "main" policy is the reference's asynchronous 1F1B-ish pipeline
+(src/train/VGG16.py:61-190): first stage keeps up to `control-count` microbatches
+in flight, draining gradients before injecting new batches; the last stage
+consumes activations, computes the loss, steps, and routes the cut-layer
+gradient back along the trace.  The reference configures middle stages but
+never implements them (src/RpcClient.py:115-118); here the genuine multi-hop
+relay stage exists.
+
+MI355X-native differences from the reference:
+* activations/gradients stay GPU-resident end to end (no .cpu().numpy());
+* by default the stage-1 autograd graph is STASHED (288 GB HBM3E makes the
+  reference's recompute-on-backward, src/train/VGG16.py:89-91, unnecessary);
+  set scheduler.recompute for reference-exact behaviour;
+* NaN detection accumulates in a device flag, synced once per round (the
+  reference syncs every batch via loss.item(), src/train/VGG16.py:168-171).
+
+Variant policies (Vanilla/Cluster_FSL/DCSL/FLEX/2LS) plug in as different
+drive loops over the same primitives (see policies.py).
+"""
+
+from __future__ import annotations
+
+import itertools
+import time
+from dataclasses import dataclass, field
+from typing import Any, Dict, Optional
+
+import torch
+
+from ..ops import functional as hf
+from .messages import ActivationMsg, GradientMsg
+
+
+def _cross_entropy(logits, labels):
+    if logits.is_cuda:
+        return hf.cross_entropy(logits, labels)
+    return torch.nn.functional.cross_entropy(logits, labels)
+
+
+@dataclass
+class StageContext:
+    client_id: int
+    layer_id: int
+    n_stages: int
+    cluster: int
+    model: Any
+    optimizer: Any
+    learning: Dict[str, Any]
+    plane: Any                      # data plane
+    control: Any                    # control plane (for PAUSE polling)
+    device: torch.device
+    train_loader: Any = None        # stage-1 only
+    # recompute=True (default) is the reference's semantics
+    # (src/train/VGG16.py:89-91): only the INPUT batch is stashed and the
+    # forward is recomputed with current weights at backward time; the first
+    # forward runs under no_grad (cheaper than the reference, which builds and
+    # discards a graph).  recompute=False stashes the live graph — only valid
+    # when control-count == 1, since the in-place optimizer step invalidates
+    # older stashed graphs.
+    recompute: bool = True
+    max_batches: Optional[int] = None
+    time_limit_s: Optional[float] = None
+    clip_grad_norm: Optional[float] = None
+    on_step: Any = None             # callback(step_idx) for benchmarking hooks
+    log_loss: Any = None            # callback(loss_tensor) optional
+
+
+def _check_pause(ctx: StageContext) -> Optional[dict]:
+    msg = ctx.control.recv(f"client_{ctx.client_id}", block=False)
+    return msg
+
+
+def train_first_stage(ctx: StageContext):
+    """Returns (result, n_samples)."""
+    model, opt, plane = ctx.model, ctx.optimizer, ctx.plane
+    cc = int(ctx.learning.get("control-count", 3))
+    batch_iter = iter(ctx.train_loader)
+    if ctx.max_batches is not None:
+        batch_iter = itertools.islice(batch_iter, ctx.max_batches)
+    inflight: Dict[int, Any] = {}
+    n_fwd = n_bwd = 0
+    data_count = 0
+    next_id = ctx.client_id * 1_000_000 + 1
+    end_data = False
+    t0 = time.monotonic()
+    model.train()
+
+    while True:
+        g = plane.recv_gradient(ctx.layer_id, ctx.client_id, block=False)
+        if g is not None:
+            stashed = inflight.pop(g.data_id)
+            opt.zero_grad()
+            if ctx.recompute:
+                out = model(stashed)          # stashed = input batch
+            else:
+                out = stashed                 # stashed = output w/ live graph
+            out.backward(gradient=g.data.to(ctx.device, non_blocking=True))
+            opt.step()
+            n_bwd += 1
+            if ctx.on_step is not None:
+                ctx.on_step(n_bwd)
+        elif not end_data and len(inflight) < cc:
+            try:
+                x, y = next(batch_iter)
+            except StopIteration:
+                end_data = True
+                continue
+            if ctx.time_limit_s and time.monotonic() - t0 > ctx.time_limit_s:
+                end_data = True
+                continue
+            x = x.to(ctx.device, non_blocking=True)
+            if ctx.recompute:
+                with torch.no_grad():
+                    out = model(x)
+            else:
+                out = model(x)
+            data_id = next_id
+            next_id += 1
+            inflight[data_id] = x if ctx.recompute else out
+            plane.send_activation(
+                ctx.layer_id, ctx.cluster,
+                ActivationMsg(data_id, out.detach(), y, [ctx.client_id]))
+            n_fwd += 1
+            data_count += x.shape[0]
+        if end_data and n_fwd == n_bwd:
+            break
+    return True, data_count
+
+
+def train_last_stage(ctx: StageContext):
+    model, opt, plane = ctx.model, ctx.optimizer, ctx.plane
+    data_count = 0
+    nan_flag = torch.zeros((), dtype=torch.bool, device=ctx.device)
+    model.train()
+    while True:
+        m = plane.recv_activation(ctx.layer_id - 1, ctx.cluster, ctx.client_id,
+                                  block=False)
+        if m is not None:
+            act = m.data.to(ctx.device, non_blocking=True).detach().requires_grad_(True)
+            labels = m.labels.to(ctx.device, non_blocking=True)
+            opt.zero_grad()
+            out = model(act)
+            loss = _cross_entropy(out, labels)
+            nan_flag |= torch.isnan(loss)
+            if ctx.log_loss is not None:
+                ctx.log_loss(loss)
+            loss.backward()
+            if ctx.clip_grad_norm:
+                torch.nn.utils.clip_grad_norm_(
+                    [p for p in model.parameters() if p.grad is not None],
+                    ctx.clip_grad_norm)
+            opt.step()
+            data_count += act.shape[0]
+            # queue key is the RECEIVER's stage (reference
+            # gradient_queue_{layer_id-1}_{to_client}, src/train/VGG16.py:43)
+            plane.send_gradient(
+                ctx.layer_id - 1, m.trace[-1],
+                GradientMsg(m.data_id, act.grad.detach(), m.trace[:-1]))
+        else:
+            msg = _check_pause(ctx)
+            if msg is not None and msg.get("action") == "PAUSE":
+                result = not bool(nan_flag.item())
+                return result, data_count
+
+
+def train_middle_stage(ctx: StageContext):
+    """Multi-hop relay: recv act -> fwd -> send down; recv grad -> bwd -> send up."""
+    model, opt, plane = ctx.model, ctx.optimizer, ctx.plane
+    cc = int(ctx.learning.get("control-count", 3))
+    inflight: Dict[int, Any] = {}
+    data_count = 0
+    model.train()
+    while True:
+        g = plane.recv_gradient(ctx.layer_id, ctx.client_id, block=False)
+        if g is not None:
+            # recompute the stage forward with current weights (same stale-
+            # weight semantics as stage 1; a stashed graph would be invalidated
+            # by the in-place optimizer steps of other in-flight microbatches)
+            act_in = inflight.pop(g.data_id)
+            act_in.requires_grad_(True)
+            opt.zero_grad()
+            out = model(act_in)
+            out.backward(gradient=g.data.to(ctx.device, non_blocking=True))
+            opt.step()
+            plane.send_gradient(
+                ctx.layer_id - 1, g.trace[-1],
+                GradientMsg(g.data_id, act_in.grad.detach(), g.trace[:-1]))
+            continue
+        if len(inflight) < cc:
+            m = plane.recv_activation(ctx.layer_id - 1, ctx.cluster, ctx.client_id,
+                                      block=False)
+            if m is not None:
+                act = m.data.to(ctx.device, non_blocking=True).detach()
+                with torch.no_grad():
+                    out = model(act)
+                inflight[m.data_id] = act
+                data_count += act.shape[0]
+                plane.send_activation(
+                    ctx.layer_id, ctx.cluster,
+                    ActivationMsg(m.data_id, out.detach(), m.labels,
+                                  m.trace + [ctx.client_id]))
+                continue
+        if not inflight:
+            msg = _check_pause(ctx)
+            if msg is not None and msg.get("action") == "PAUSE":
+                return True, data_count
+
+
+def run_stage(ctx: StageContext):
+    if ctx.layer_id == 1:
+        return train_first_stage(ctx)
+    if ctx.layer_id == ctx.n_stages:
+        return train_last_stage(ctx)
+    return train_middle_stage(ctx)
