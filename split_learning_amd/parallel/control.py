@@ -64,6 +64,8 @@ class StoreControl(ControlPlane):
     def __init__(self, store):
         self.store = store
         self._read: Dict[str, int] = {}
+        # one TCPStore object = one socket; serialise all ops on it (the
+        # underlying connection is not safe under concurrent threads)
         self._lock = threading.Lock()
 
     @staticmethod
@@ -74,22 +76,42 @@ class StoreControl(ControlPlane):
                               wait_for_workers=False)
         return StoreControl(store)
 
+    # TCPStore rejects payloads > 8 MiB; chunk large blobs (UPDATE messages
+    # carry full partition state dicts).
+    CHUNK = 4 * 1024 * 1024
+
     def send(self, dst, msg):
         blob = pickle.dumps(msg, protocol=pickle.HIGHEST_PROTOCOL)
         with self._lock:
-            seq = self.store.add(f"mqc_w_{dst}", 1)  # 1-based write cursor
-        self.store.set(f"mq_{dst}_{seq - 1}", blob)
+            seq = self.store.add(f"mqc_w_{dst}", 1) - 1  # 0-based slot
+            if len(blob) <= self.CHUNK:
+                self.store.set(f"mq_{dst}_{seq}", blob)
+            else:
+                n = (len(blob) + self.CHUNK - 1) // self.CHUNK
+                for i in range(n):
+                    self.store.set(f"mq_{dst}_{seq}_c{i}",
+                                   blob[i * self.CHUNK:(i + 1) * self.CHUNK])
+                self.store.set(f"mq_{dst}_{seq}", b"__CHUNKED__:%d" % n)
 
     def recv(self, dst, block=True, timeout=None):
         deadline = None if timeout is None else time.monotonic() + timeout
         while True:
             nxt = self._read.get(dst, 0)
-            avail = self.store.add(f"mqc_w_{dst}", 0)
-            if avail > nxt:
-                blob = self.store.get(f"mq_{dst}_{nxt}")
-                self.store.delete_key(f"mq_{dst}_{nxt}")
-                self._read[dst] = nxt + 1
-                return pickle.loads(blob)
+            with self._lock:
+                avail = self.store.add(f"mqc_w_{dst}", 0)
+                if avail > nxt:
+                    blob = self.store.get(f"mq_{dst}_{nxt}")
+                    self.store.delete_key(f"mq_{dst}_{nxt}")
+                    if blob.startswith(b"__CHUNKED__:"):
+                        n = int(blob.split(b":", 1)[1])
+                        parts = []
+                        for i in range(n):
+                            k = f"mq_{dst}_{nxt}_c{i}"
+                            parts.append(bytes(self.store.get(k)))
+                            self.store.delete_key(k)
+                        blob = b"".join(parts)
+                    self._read[dst] = nxt + 1
+                    return pickle.loads(bytes(blob))
             if not block:
                 return None
             if deadline is not None and time.monotonic() > deadline:

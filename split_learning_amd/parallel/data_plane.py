@@ -8,12 +8,22 @@
   the reference's pickled-numpy RabbitMQ queues (src/train/VGG16.py:20-53):
   no host round-trip — tensors leave and arrive GPU-resident.
 
+Deadlock discipline (RCCL kernels spin until the peer joins, and all ops of
+one communicator execute in enqueue order on its internal stream):
+  1. forward (activation) and backward (gradient) traffic use SEPARATE
+     process groups -> separate RCCL comms/streams, so cross-direction
+     enqueue-order cycles cannot form;
+  2. consumers PRE-POST a ring of `depth` irecvs per edge (depth >= the
+     pipeline's control-count), so producers can run ahead;
+  3. at STOP, producers send one poison message per owned edge so outstanding
+     pre-posted irecvs complete before process-group teardown.
+
 Routing: the reference's shared per-cluster AMQP queue gives multi-consumer
 work-stealing for free; p2p is pairwise, so the server assigns static
 round-robin producer->consumer edges (the DCSL variant's targeted routing,
 other/DCSL/src/Scheduler.py:110-115, is the template) and gradient edges are
 the reverse.  Each message is a (header, labels, payload) tensor triple with
-static shapes so receivers can pre-post irecvs.
+static shapes so receivers can pre-post.
 """
 
 from __future__ import annotations
@@ -45,8 +55,6 @@ class LoopbackData:
                 table[key] = queue.Queue()
             return table[key]
 
-    # activation queue is shared per (producer_stage, cluster) like the
-    # reference's intermediate_queue_{layer}_{cluster}
     def send_activation(self, stage: int, cluster: int, msg: ActivationMsg,
                         dst_client: Optional[int] = None):
         self._q(self._act, (stage, cluster)).put(msg)
@@ -74,85 +82,142 @@ class LoopbackData:
         pass
 
 
-class _Edge:
-    """One directed p2p channel with static shapes; keeps an irecv pipelined."""
+class RecvRing:
+    """Pre-posted irecv ring for one (peer, shape) channel.
 
-    def __init__(self, peer: int, shape, batch: int, device, tag_base: int,
+    Message k is posted/sent with tags (3k, 3k+1, 3k+2) for its
+    header/labels/payload: gloo matches concurrent p2p ops by tag (same-tag
+    outstanding recvs match nondeterministically), while RCCL ignores tags but
+    matches strictly FIFO per pair — so sequence tags make BOTH backends
+    deterministic."""
+
+    def __init__(self, peer: int, shape, batch: int, device, group, depth: int,
                  dtype=torch.float32):
         self.peer = peer
-        self.device = device
-        self.header = torch.zeros(3 + MAX_TRACE, dtype=torch.int64, device=device)
-        self.labels = torch.zeros(batch, dtype=torch.int64, device=device)
-        self.payload = torch.zeros(*shape, dtype=dtype, device=device)
-        self.works = None
+        self.group = group
+        self.seq = 0
+        self.depth = depth
+        # gloo's Work.is_completed() never reflects async recv completion
+        # (observed: buffers filled, flag stays False), so on gloo a helper
+        # thread blocking-waits each message and feeds a python queue; on
+        # nccl/RCCL is_completed() is CUDA-event-backed and polling works.
+        self.threaded = dist.get_backend(group) == "gloo" if group is not None \
+            else dist.get_backend() == "gloo"
+        self.slots = []
+        for _ in range(depth):
+            self.slots.append({
+                "header": torch.zeros(3 + MAX_TRACE, dtype=torch.int64, device=device),
+                "labels": torch.zeros(batch, dtype=torch.int64, device=device),
+                "payload": torch.zeros(*shape, dtype=dtype, device=device),
+                "works": None,
+            })
+        self.order = collections.deque()
+        for i in range(depth):
+            self._post(i)
+        if self.threaded:
+            self._q: queue.Queue = queue.Queue()
+            self._stop = False
+            self._thread = threading.Thread(target=self._pump, daemon=True)
+            self._thread.start()
 
-    def post_recv(self):
-        w1 = dist.irecv(self.header, src=self.peer)
-        w2 = dist.irecv(self.labels, src=self.peer)
-        w3 = dist.irecv(self.payload, src=self.peer)
-        self.works = (w1, w2, w3)
+    def _post(self, i):
+        s = self.slots[i]
+        t = self.seq * 3
+        self.seq += 1
+        w1 = dist.irecv(s["header"], src=self.peer, group=self.group, tag=t)
+        w2 = dist.irecv(s["labels"], src=self.peer, group=self.group, tag=t + 1)
+        w3 = dist.irecv(s["payload"], src=self.peer, group=self.group, tag=t + 2)
+        s["works"] = (w1, w2, w3)
+        self.order.append(i)
+
+    def _take(self):
+        """Blocking-wait the oldest posted message; no repost."""
+        i = self.order[0]
+        s = self.slots[i]
+        for w in s["works"]:
+            w.wait()
+        out = (s["header"].clone(), s["labels"].clone(), s["payload"].clone())
+        self.order.popleft()
+        return i, out
+
+    def _consume(self):
+        i, out = self._take()
+        if int(out[0][0]) != -1:  # poison messages are never reposted
+            self._post(i)
+        return out
+
+    def _pump(self):
+        while not self._stop:
+            triple = self._consume()
+            if int(triple[0][0]) == -1:
+                self._q.put(None)
+                return
+            self._q.put(triple)
 
     def poll(self):
-        """Return (header, labels, payload) clones if a message landed, else None."""
-        if self.works is None:
-            self.post_recv()
-        if not self.works[0].is_completed():
+        if self.threaded:
+            try:
+                return self._q.get_nowait()
+            except queue.Empty:
+                return None
+        i = self.order[0]
+        if not all(w.is_completed() for w in self.slots[i]["works"]):
             return None
-        for w in self.works:
-            w.wait()
-        out = (self.header.clone(), self.labels.clone(), self.payload.clone())
-        self.works = None
-        self.post_recv()
-        return out
+        return self._consume()
 
     def wait(self):
-        if self.works is None:
-            self.post_recv()
-        for w in self.works:
-            w.wait()
-        out = (self.header.clone(), self.labels.clone(), self.payload.clone())
-        self.works = None
-        self.post_recv()
-        return out
+        if self.threaded:
+            return self._q.get(block=True)
+        return self._consume()
+
+    def drain(self):
+        """Consume every outstanding posted recv (the peer sends exactly
+        `depth` poison messages at shutdown, so this terminates)."""
+        if self.threaded:
+            self._stop = True
+            self._thread.join(timeout=60.0)
+        while self.order:
+            self._take()
 
 
 class P2PData:
-    """torch.distributed p2p data plane (RCCL on GPU, gloo on CPU).
-
-    Construction needs the routing plan the server computed:
-      * down_peer: rank to send activations to (None for last stage)
-      * up_peers: ranks this stage receives activations from
-      * act_shape_out / act_shape_in: payload shapes (batch-major)
-      * grad peers mirror activation edges in reverse.
-    """
+    """torch.distributed p2p data plane (RCCL on GPU, gloo on CPU)."""
 
     def __init__(self, my_rank: int, device: torch.device, batch: int,
                  down_peer: Optional[int], up_peers: List[int],
-                 act_shape_out, act_shape_in, grad_from_down: bool):
+                 act_shape_out, act_shape_in, grad_from_down: bool,
+                 group_fwd=None, group_bwd=None, depth: int = 4):
         self.rank = my_rank
         self.device = device
         self.batch = batch
         self.down_peer = down_peer
         self.up_peers = list(up_peers)
+        self.group_fwd = group_fwd
+        self.group_bwd = group_bwd
         self._pending_sends = collections.deque()
-        # recv edges for activations (from each upstream peer)
-        self._act_edges = {p: _Edge(p, act_shape_in, batch, device, 0)
+        self._send_seq: Dict[tuple, int] = {}
+        self._act_rings = {p: RecvRing(p, act_shape_in, batch, device, group_fwd,
+                                       depth)
                            for p in self.up_peers} if act_shape_in else {}
-        # recv edge for gradients (from the downstream peer, same shape as out act)
-        self._grad_edge = (_Edge(down_peer, act_shape_out, batch, device, 1)
+        self._grad_ring = (RecvRing(down_peer, act_shape_out, batch, device,
+                                    group_bwd, depth)
                            if (grad_from_down and down_peer is not None) else None)
         self._rr = 0
 
-    # -- helpers -----------------------------------------------------------
+    # -- send helpers -------------------------------------------------------
     def _reap(self):
         while self._pending_sends and all(w.is_completed()
                                           for w in self._pending_sends[0][0]):
             self._pending_sends.popleft()
 
-    def _send_triple(self, dst: int, header, labels, payload):
-        w1 = dist.isend(header, dst=dst)
-        w2 = dist.isend(labels, dst=dst)
-        w3 = dist.isend(payload, dst=dst)
+    def _send_triple(self, dst, header, labels, payload, group):
+        key = (dst, id(group))
+        seq = self._send_seq.get(key, 0)
+        self._send_seq[key] = seq + 1
+        t = seq * 3
+        w1 = dist.isend(header, dst=dst, group=group, tag=t)
+        w2 = dist.isend(labels, dst=dst, group=group, tag=t + 1)
+        w3 = dist.isend(payload, dst=dst, group=group, tag=t + 2)
         self._pending_sends.append(((w1, w2, w3), (header, labels, payload)))
         self._reap()
 
@@ -168,12 +233,9 @@ class P2PData:
     @staticmethod
     def _unpack_header(h: torch.Tensor):
         hc = h.cpu()
-        data_id = int(hc[0])
-        tlen = int(hc[2])
-        trace = [int(hc[3 + i]) for i in range(tlen)]
-        return data_id, trace
+        return int(hc[0]), [int(hc[3 + i]) for i in range(int(hc[2]))]
 
-    # -- activation / gradient API (mirrors LoopbackData) ------------------
+    # -- activation / gradient API (mirrors LoopbackData) -------------------
     def send_activation(self, stage: int, cluster: int, msg: ActivationMsg,
                         dst_client: Optional[int] = None):
         dst = self.down_peer if dst_client is None else dst_client
@@ -182,50 +244,71 @@ class P2PData:
                   else torch.zeros(self.batch, dtype=torch.int64))
         labels = labels.to(self.device, non_blocking=True)
         payload = msg.data.to(self.device).contiguous()
-        self._send_triple(dst, header, labels, payload)
+        self._send_triple(dst, header, labels, payload, self.group_fwd)
 
     def recv_activation(self, prev_stage: int, cluster: int, my_client: int,
                         block=False, timeout=None) -> Optional[ActivationMsg]:
         n = len(self.up_peers)
         if n == 0:
             return None
-        for i in range(n):
-            peer = self.up_peers[(self._rr + i) % n]
-            got = self._act_edges[peer].poll()
-            if got is not None:
-                self._rr = (self._rr + i + 1) % n
-                return self._to_act(got)
-        if block:
-            # single-peer blocking wait; multi-peer keeps polling
-            if n == 1:
-                return self._to_act(self._act_edges[self.up_peers[0]].wait())
-            while True:
-                for peer in self.up_peers:
-                    got = self._act_edges[peer].poll()
-                    if got is not None:
-                        return self._to_act(got)
-        return None
-
-    def _to_act(self, triple) -> ActivationMsg:
-        header, labels, payload = triple
-        data_id, trace = self._unpack_header(header)
-        return ActivationMsg(data_id, payload, labels, trace)
+        while True:
+            for i in range(n):
+                peer = self.up_peers[(self._rr + i) % n]
+                got = self._act_rings[peer].poll()
+                if got is not None:
+                    self._rr = (self._rr + i + 1) % n
+                    header, labels, payload = got
+                    data_id, trace = self._unpack_header(header)
+                    return ActivationMsg(data_id, payload, labels, trace)
+            if not block:
+                return None
 
     def send_gradient(self, stage: int, to_client: int, msg: GradientMsg):
         header = self._pack_header(msg.data_id, msg.trace)
         labels = torch.zeros(self.batch, dtype=torch.int64, device=self.device)
-        self._send_triple(to_client, header, labels, msg.data.contiguous())
+        self._send_triple(to_client, header, labels, msg.data.contiguous(),
+                          self.group_bwd)
 
     def recv_gradient(self, stage: int, client: int, block=False,
                       timeout=None) -> Optional[GradientMsg]:
-        if self._grad_edge is None:
+        if self._grad_ring is None:
             return None
-        got = self._grad_edge.wait() if block else self._grad_edge.poll()
+        got = self._grad_ring.wait() if block else self._grad_ring.poll()
         if got is None:
             return None
         header, _, payload = got
         data_id, trace = self._unpack_header(header)
         return GradientMsg(data_id, payload, trace)
+
+    # -- shutdown -----------------------------------------------------------
+    def poison_shutdown(self):
+        """Satisfy every pre-posted irecv: each sender pushes `depth` dummy
+        messages per owned edge, then receivers drain."""
+        if self.down_peer is not None and self._grad_ring is not None:
+            depth = len(self._grad_ring.slots)
+        else:
+            depth = 4
+        if self.down_peer is not None:
+            # I produce activations for down_peer's act ring
+            for _ in range(depth):
+                self.send_activation(0, 0, ActivationMsg(
+                    -1, torch.zeros(*self._grad_ring.slots[0]["payload"].shape,
+                                    device=self.device)
+                    if self._grad_ring else torch.zeros(1, device=self.device),
+                    None, []))
+        for peer in self.up_peers:
+            # I produce gradients for peer's grad ring
+            ring = self._act_rings[peer]
+            for _ in range(len(ring.slots)):
+                self._send_triple(
+                    peer, self._pack_header(-1, []),
+                    torch.zeros(self.batch, dtype=torch.int64, device=self.device),
+                    torch.zeros_like(ring.slots[0]["payload"]), self.group_bwd)
+        self.flush()
+        for ring in self._act_rings.values():
+            ring.drain()
+        if self._grad_ring is not None:
+            self._grad_ring.drain()
 
     def flush(self):
         for works, _bufs in self._pending_sends:

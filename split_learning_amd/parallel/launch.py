@@ -79,7 +79,18 @@ def run_loopback(config: Dict[str, Any], device: str = "cpu",
     return server, runtimes
 
 
-def make_p2p_plane_factory(rank: int, device: torch.device):
+def make_p2p_groups():
+    """Create the forward/backward process groups (collective: every rank must
+    call this once, in the same order, right after init_process_group)."""
+    import torch.distributed as dist
+    world = list(range(dist.get_world_size()))
+    group_fwd = dist.new_group(world)
+    group_bwd = dist.new_group(world)
+    return group_fwd, group_bwd
+
+
+def make_p2p_plane_factory(rank: int, device: torch.device, group_fwd, group_bwd,
+                           depth: int = 4):
     """plane_factory for ClientRuntime in one-process-per-GPU mode: builds a
     P2PData from the routing dict the server ships in START."""
 
@@ -90,6 +101,62 @@ def make_p2p_plane_factory(rank: int, device: torch.device):
             act_shape_out=routing["act_shape_out"],
             act_shape_in=routing["act_shape_in"],
             grad_from_down=routing["act_shape_out"] is not None,
+            group_fwd=group_fwd, group_bwd=group_bwd, depth=depth,
         )
 
     return factory
+
+
+def run_p2p_client(config: Dict[str, Any], rank: int, world: int,
+                   device: torch.device, store_addr: str, store_port: int,
+                   max_batches: Optional[int] = None, on_step=None,
+                   checkpoint_dir: str = ".", logger=None):
+    """One-process-per-GPU worker: rank 0 also runs the server in a thread.
+
+    Stage assignment: ranks are stage-major in registration order — the first
+    clients[0] ranks are stage 1, the next clients[1] stage 2, etc.
+    client_id == rank (the p2p routing the server computes uses client ids as
+    ranks).
+    """
+    from .control import StoreControl
+
+    clients_per_stage = config["server"]["clients"]
+    assert sum(clients_per_stage) == world, \
+        f"sum(clients)={sum(clients_per_stage)} must equal world size {world}"
+    layer_id = None
+    acc = 0
+    for stage, n in enumerate(clients_per_stage, start=1):
+        if rank < acc + n:
+            layer_id = stage
+            break
+        acc += n
+    group_fwd, group_bwd = make_p2p_groups()
+    control = StoreControl.create(store_addr, store_port, is_server=(rank == 0))
+    logger = logger or Logger(f"{config['log_path']}/app.log", config["debug_mode"])
+
+    server_thread = None
+    server = None
+    if rank == 0:
+        # the server thread gets its OWN store connection (a TCPStore object is
+        # a single socket and must not be shared across threads)
+        server_control = StoreControl.create(store_addr, store_port,
+                                             is_server=False)
+        server = Server(config, server_control, logger=logger,
+                        checkpoint_dir=checkpoint_dir)
+        server_thread = threading.Thread(target=server.run, daemon=True)
+        server_thread.start()
+
+    rt = ClientRuntime(
+        rank, layer_id, control, None, device, logger=logger,
+        scheduler_cfg=config.get("scheduler"),
+        plane_factory=make_p2p_plane_factory(
+            rank, device, group_fwd, group_bwd,
+            depth=max(2, int(config["learning"].get("control-count", 3)) + 1)))
+    rt.register()
+    rt.run(max_batches=max_batches, on_step=on_step)
+    if server_thread is not None:
+        server_thread.join(timeout=120.0)
+    import torch.distributed as dist
+    if dist.is_initialized():
+        dist.barrier()  # keep rank 0's master store alive until all ranks finish
+    return server, rt

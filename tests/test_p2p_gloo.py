@@ -1,0 +1,73 @@
+"""Multi-process p2p transport tests on CPU (gloo backend, world_size 2/3) —
+the same P2PData/StoreControl path that runs RCCL-over-xGMI on the GPU node,
+proving the distributed round protocol end to end without a GPU."""
+
+import os
+import socket
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _cfg(tmpdir, clients, cuts, num_sample=48, batch=16):
+    from split_learning_amd.config import load_config
+    return load_config(None, overrides={
+        "server": {
+            "global-round": 1, "clients": clients, "model": "VGG16",
+            "data-name": "CIFAR10", "validation": False,
+            "parameters": {"load": True, "save": True},
+            "data-distribution": {"num-sample": num_sample, "num-label": 10,
+                                  "non-iid": False, "dirichlet": {"alpha": 1},
+                                  "refresh": True},
+            "manual": {"cluster-mode": False, "no-cluster": {"cut-layers": cuts}},
+        },
+        "log_path": str(tmpdir), "debug_mode": False,
+        "learning": {"batch-size": batch, "control-count": 3,
+                     "learning-rate": 5e-4, "momentum": 0.5,
+                     "weight-decay": 0.01},
+    })
+
+
+def _worker(rank, world, pg_port, ctl_port, tmpdir, clients, cuts):
+    import torch.distributed as dist
+    from split_learning_amd.parallel.launch import run_p2p_client
+    dist.init_process_group("gloo", init_method=f"tcp://127.0.0.1:{pg_port}",
+                            rank=rank, world_size=world)
+    cfg = _cfg(tmpdir, clients, cuts)
+    run_p2p_client(cfg, rank, world, torch.device("cpu"), "127.0.0.1", ctl_port,
+                   checkpoint_dir=str(tmpdir))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_p2p_two_rank_round(tmp_path):
+    world = 2
+    pg_port, ctl_port = _free_port(), _free_port()
+    mp.spawn(_worker, args=(world, pg_port, ctl_port, str(tmp_path), [1, 1], [7]),
+             nprocs=world, join=True)
+    ckpt = os.path.join(str(tmp_path), "VGG16_CIFAR10.pth")
+    assert os.path.exists(ckpt)
+    sd = torch.load(ckpt, weights_only=True)
+    from split_learning_amd.models import get_model_class
+    assert set(sd.keys()) == set(get_model_class("VGG16", "CIFAR10")().state_dict().keys())
+
+
+@pytest.mark.timeout(900)
+def test_p2p_three_stage_round(tmp_path):
+    """3 ranks, cuts [7,14]: middle-stage relay over p2p."""
+    world = 3
+    pg_port, ctl_port = _free_port(), _free_port()
+    mp.spawn(_worker, args=(world, pg_port, ctl_port, str(tmp_path), [1, 1, 1],
+                            [7, 14]),
+             nprocs=world, join=True)
+    assert os.path.exists(os.path.join(str(tmp_path), "VGG16_CIFAR10.pth"))
