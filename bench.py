@@ -89,56 +89,78 @@ def run_colocated(device, steps):
     return nan_flag
 
 
-def run_distributed(rank, world, device, steps):
-    """One rank per GPU; pair (r, r + world/2) forms a pipeline."""
-    import torch.distributed as dist
-    from split_learning_amd.ops import functional as hf
+class DistPipeline:
+    """One rank per GPU; pair (r, r + world/2) forms a pipeline.  Uses the
+    production P2PData plane (pre-posted irecv rings, fwd/bwd on separate
+    RCCL communicators) so the bench measures the real engine transport."""
 
-    half = world // 2
-    is_first = rank < half
-    peer = rank + half if is_first else rank - half
-    act_shape = (BATCH, 64, 16, 16)  # VGG16 cut=7 boundary ([B,64,16,16])
+    def __init__(self, rank, world, device):
+        from split_learning_amd.parallel.data_plane import P2PData
+        from split_learning_amd.parallel.launch import make_p2p_groups
+        half = world // 2
+        self.rank = rank
+        self.device = device
+        self.is_first = rank < half
+        self.peer = rank + half if self.is_first else rank - half
+        act_shape = (BATCH, 64, 16, 16)  # VGG16 cut=7 boundary
+        gf, gb = make_p2p_groups()
+        if self.is_first:
+            self.model, self.opt = build_stage([0, CUT], device)
+            self.plane = P2PData(rank, device, BATCH, down_peer=self.peer,
+                                 up_peers=[], act_shape_out=act_shape,
+                                 act_shape_in=None, grad_from_down=True,
+                                 group_fwd=gf, group_bwd=gb,
+                                 depth=CONTROL_COUNT + 1)
+        else:
+            self.model, self.opt = build_stage([CUT, -1], device)
+            self.plane = P2PData(rank, device, BATCH, down_peer=None,
+                                 up_peers=[self.peer], act_shape_out=None,
+                                 act_shape_in=act_shape, grad_from_down=False,
+                                 group_fwd=gf, group_bwd=gb,
+                                 depth=CONTROL_COUNT + 1)
+        self.next_id = 1
 
-    if is_first:
-        model, opt = build_stage([0, CUT], device)
-        xs, ys = make_batches(device, steps, seed=rank)
-        grad_buf = torch.zeros(act_shape, device=device)
-        inflight = []
-        for i in range(steps):
-            x, y = xs[i], ys[i]
-            with torch.no_grad():
-                act = s_out = model(x)
-            dist.send(act.contiguous(), dst=peer)
-            dist.send(y, dst=peer)
-            inflight.append(x)
-            if len(inflight) >= CONTROL_COUNT or i == steps - 1:
-                while inflight:
-                    dist.recv(grad_buf, src=peer)
-                    xo = inflight.pop(0)
-                    opt.zero_grad()
-                    out = model(xo)
-                    out.backward(gradient=grad_buf)
-                    opt.step()
-                    if len(inflight) < CONTROL_COUNT - 1 and i < steps - 1:
-                        break
-        return torch.zeros((), dtype=torch.bool, device=device)
-    else:
-        model, opt = build_stage([CUT, -1], device)
-        act_buf = torch.zeros(act_shape, device=device)
-        y_buf = torch.zeros(BATCH, dtype=torch.int64, device=device)
-        nan_flag = torch.zeros((), dtype=torch.bool, device=device)
-        for i in range(steps):
-            dist.recv(act_buf, src=peer)
-            dist.recv(y_buf, src=peer)
-            act = act_buf.clone().requires_grad_(True)
-            opt.zero_grad()
-            logits = model(act)
-            loss = hf.cross_entropy(logits, y_buf) if logits.is_cuda else \
-                torch.nn.functional.cross_entropy(logits, y_buf)
-            nan_flag |= torch.isnan(loss)
-            loss.backward()
-            opt.step()
-            dist.send(act.grad.contiguous(), dst=peer)
+    def run(self, steps):
+        from split_learning_amd.ops import functional as hf
+        from split_learning_amd.parallel.messages import ActivationMsg, GradientMsg
+        import collections
+        nan_flag = torch.zeros((), dtype=torch.bool, device=self.device)
+        if self.is_first:
+            xs, ys = make_batches(self.device, steps, seed=self.rank + self.next_id)
+            inflight = collections.deque()
+
+            def bwd_one():
+                g = self.plane.recv_gradient(1, self.rank, block=True)
+                xo = inflight.popleft()
+                self.opt.zero_grad()
+                out = self.model(xo)
+                out.backward(gradient=g.data)
+                self.opt.step()
+
+            for i in range(steps):
+                with torch.no_grad():
+                    act = self.model(xs[i])
+                self.plane.send_activation(1, 0, ActivationMsg(
+                    self.next_id, act, ys[i], [self.rank]))
+                self.next_id += 1
+                inflight.append(xs[i])
+                if len(inflight) >= CONTROL_COUNT:
+                    bwd_one()
+            while inflight:
+                bwd_one()
+        else:
+            for i in range(steps):
+                m = self.plane.recv_activation(1, 0, self.rank, block=True)
+                act = m.data.requires_grad_(True)
+                self.opt.zero_grad()
+                logits = self.model(act)
+                loss = hf.cross_entropy(logits, m.labels) if logits.is_cuda else \
+                    torch.nn.functional.cross_entropy(logits, m.labels)
+                nan_flag |= torch.isnan(loss)
+                loss.backward()
+                self.opt.step()
+                self.plane.send_gradient(1, self.peer, GradientMsg(
+                    m.data_id, act.grad.detach(), []))
         return nan_flag
 
 
@@ -181,8 +203,11 @@ def main():
             if have_gpu:
                 torch.cuda.synchronize()
 
-    runner = (lambda n: run_distributed(rank, world, device, n)) if dist_mode \
-        else (lambda n: run_colocated(device, n))
+    if dist_mode:
+        pipeline = DistPipeline(rank, world, device)
+        runner = pipeline.run
+    else:
+        runner = lambda n: run_colocated(device, n)  # noqa: E731
 
     log(f"[bench] warmup {args.warmup} steps (rank {rank}/{world}, {device})")
     nan_w = runner(args.warmup)
@@ -231,6 +256,8 @@ def main():
 
     if dist_mode:
         import torch.distributed as dist
+        pipeline.plane.poison_shutdown()
+        dist.barrier()
         dist.destroy_process_group()
 
 
