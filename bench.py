@@ -56,37 +56,75 @@ def make_batches(device, n, seed=0):
     return x, y
 
 
-def run_colocated(device, steps):
-    """Both stages on one GPU, serial in-process pipeline (N=1 path)."""
-    from split_learning_amd.ops import functional as hf
-    s1_model, s1_opt = build_stage([0, CUT], device)
-    s2_model, s2_opt = build_stage([CUT, -1], device)
-    xs, ys = make_batches(device, steps)
-    nan_flag = torch.zeros((), dtype=torch.bool, device=device)
+class ColocatedPipeline:
+    """Both stages on one GPU (N=1 path).  On GPU the whole microbatch step —
+    stage-1 fwd, stage-2 fwd+loss+bwd+step, stage-1 recompute+bwd+step — is
+    captured once into a hipGraph and replayed per step (CDNA guide: capture
+    launch-bound inner loops in hipGraphs; dropout offsets advance via a
+    device-side counter so masks stay fresh across replays)."""
 
-    def ce(logits, labels):
+    def __init__(self, device, use_graphs=True):
+        from split_learning_amd.ops import functional as hf
+        self.hf = hf
+        self.device = device
+        self.s1_model, self.s1_opt = build_stage([0, CUT], device)
+        self.s2_model, self.s2_opt = build_stage([CUT, -1], device)
+        self.x_buf = torch.zeros(BATCH, 3, 32, 32, device=device)
+        self.y_buf = torch.zeros(BATCH, dtype=torch.int64, device=device)
+        self.nan_flag = torch.zeros((), dtype=torch.bool, device=device)
+        self.use_graphs = use_graphs and device.type == "cuda"
+        self.graph = None
+        self.batch_seed = 0
+
+    def _ce(self, logits, labels):
         if logits.is_cuda:
-            return hf.cross_entropy(logits, labels)
+            return self.hf.cross_entropy(logits, labels)
         return torch.nn.functional.cross_entropy(logits, labels)
 
-    for i in range(steps):
-        x, y = xs[i], ys[i]
+    def _step(self):
         with torch.no_grad():
-            act = s1_model(x)
-        # stage 2: fwd + loss + bwd + step
+            act = self.s1_model(self.x_buf)
         act_in = act.detach().requires_grad_(True)
-        s2_opt.zero_grad()
-        logits = s2_model(act_in)
-        loss = ce(logits, y)
-        nan_flag |= torch.isnan(loss)
+        self.s2_opt.zero_grad()
+        logits = self.s2_model(act_in)
+        loss = self._ce(logits, self.y_buf)
+        self.nan_flag |= torch.isnan(loss)
         loss.backward()
-        s2_opt.step()
-        # stage 1: recompute fwd with grad, bwd from cut gradient, step
-        s1_opt.zero_grad()
-        out = s1_model(x)
+        self.s2_opt.step()
+        self.s1_opt.zero_grad()
+        out = self.s1_model(self.x_buf)
         out.backward(gradient=act_in.grad)
-        s1_opt.step()
-    return nan_flag
+        self.s1_opt.step()
+
+    def _capture(self):
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                self._step()
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self._step()
+
+    def run(self, steps):
+        self.batch_seed += 1
+        xs, ys = make_batches(self.device, steps, seed=self.batch_seed)
+        if self.use_graphs and self.graph is None:
+            try:
+                self._capture()
+                log("[bench] hipGraph capture of the full step: ok")
+            except Exception as e:  # noqa: BLE001
+                log(f"[bench] graph capture failed ({e!r}); eager fallback")
+                self.use_graphs = False
+        for i in range(steps):
+            self.x_buf.copy_(xs[i])
+            self.y_buf.copy_(ys[i])
+            if self.graph is not None:
+                self.graph.replay()
+            else:
+                self._step()
+        return self.nan_flag
 
 
 class DistPipeline:
@@ -170,6 +208,8 @@ def main():
     ap.add_argument("--steps", type=int, default=64)
     ap.add_argument("--warmup", type=int, default=16)
     ap.add_argument("--device", default=None)
+    ap.add_argument("--no-graphs", action="store_true",
+                    help="disable hipGraph step capture (N=1 path)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -205,9 +245,9 @@ def main():
 
     if dist_mode:
         pipeline = DistPipeline(rank, world, device)
-        runner = pipeline.run
     else:
-        runner = lambda n: run_colocated(device, n)  # noqa: E731
+        pipeline = ColocatedPipeline(device, use_graphs=not args.no_graphs)
+    runner = pipeline.run
 
     log(f"[bench] warmup {args.warmup} steps (rank {rank}/{world}, {device})")
     nan_w = runner(args.warmup)

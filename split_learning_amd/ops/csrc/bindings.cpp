@@ -36,6 +36,8 @@ at::Tensor gelu_bwd(const at::Tensor&, const at::Tensor&);
 at::Tensor tanh_fwd(const at::Tensor&);
 at::Tensor tanh_bwd(const at::Tensor&, const at::Tensor&);
 std::vector<at::Tensor> dropout_fwd(const at::Tensor&, double, int64_t, int64_t);
+std::vector<at::Tensor> dropout_fwd_dev(const at::Tensor&, double, int64_t,
+                                        const at::Tensor&);
 at::Tensor dropout_bwd(const at::Tensor&, const at::Tensor&, double);
 std::vector<at::Tensor> maxpool2x2_fwd(const at::Tensor&);
 at::Tensor maxpool2x2_bwd(const at::Tensor&, const at::Tensor&, int, int);
@@ -78,6 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tanh_fwd", &slk::tanh_fwd);
   m.def("tanh_bwd", &slk::tanh_bwd);
   m.def("dropout_fwd", &slk::dropout_fwd);
+  m.def("dropout_fwd_dev", &slk::dropout_fwd_dev);
   m.def("dropout_bwd", &slk::dropout_bwd);
   m.def("maxpool2x2_fwd", &slk::maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &slk::maxpool2x2_bwd);

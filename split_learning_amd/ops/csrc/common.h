@@ -37,6 +37,24 @@ __device__ __forceinline__ float slk_uniform(uint64_t seed, uint64_t offset,
   return (float)(h >> 40) * (1.0f / 16777216.0f);
 }
 
+// Fast integer division by a runtime constant (one 64-bit mul + shift).
+// Valid for dividend < 2^24 and divisor < 2^16 (all tensor-index math here):
+// m = floor(2^40/d)+1; q = (n*m) >> 40 == n/d exactly when n*d < 2^40.
+struct FastDiv {
+  unsigned d;
+  unsigned long long m;
+  void init(unsigned d_) {
+    d = d_ == 0 ? 1 : d_;
+    m = (0x10000000000ull / d) + 1;
+  }
+  __device__ __forceinline__ unsigned div(unsigned n) const {
+    return (unsigned)(((unsigned long long)n * m) >> 40);
+  }
+  __device__ __forceinline__ unsigned mod(unsigned n, unsigned q) const {
+    return n - q * d;
+  }
+};
+
 // block-level reduction into lane 0 of wave 0 (sums `val` over blockDim.x
 // threads; blockDim.x must be a multiple of 64 and <= 1024)
 template <typename T>

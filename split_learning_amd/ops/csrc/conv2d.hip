@@ -2,12 +2,19 @@
 //
 // Forward:     y[Co, (B,OH,OW)] = W[Co, (Ci,KH,KW)] @ im2col(x)
 // Bwd-data:    gx[Ci, (B,H,W)]  = W^T-gather @ col(gy)   (stride-aware)
-// Bwd-weight:  gw[Co, (Ci,KH,KW)] = gy-gather @ im2col(x)^T, split-K over
-//              (B,OH,OW) with atomic accumulation (small-spatial tails need it:
-//              SURVEY.md §7 hard-part 1).
-// Covers every conv in the model zoo: 3x3 s1/s2 p1, 1x1, and ViT's 4x4 s4
-// patch embed (reference conv sites: src/model/VGG16_CIFAR10.py:10-94,
-// other/Vanilla_SL/src/model/MobileNetv1_CIFAR10.py, ViT_CIFAR10.py:44).
+// Bwd-weight:  gw[Co, (Ci,KH,KW)] = gy-gather @ im2col(x)^T
+//
+// Performance notes (profiled on MI355X, profiles/):
+// * gather index math is the hot cost at these shapes, so the kernel geometry
+//   (KH, KW, stride, pad) is a TEMPLATE specialisation for the model zoo's
+//   cases — 3x3 s1 p1, 3x3 s2 p1, 1x1, 4x4 s4 (ViT patch) — making every
+//   division strength-reduced; runtime dims (OW, OH*OW, H*W) divide through
+//   FastDiv magics (common.h);
+// * every conv GEMM split-Ks when its tile grid underfills the 256 CUs
+//   (deep VGG layers have N = B*OH*OW as small as 128), with fp32 atomic
+//   accumulation and first-split-gated bias;
+// * covers every conv in the zoo (reference conv sites:
+//   src/model/VGG16_CIFAR10.py:10-94, MobileNetv1, ViT patch embed).
 #include <torch/extension.h>
 #include <ATen/ATen.h>
 #include <c10/hip/HIPStream.h>
@@ -18,28 +25,48 @@ namespace slk {
 
 struct ConvGeom {
   int B, Ci, H, W, Co, KH, KW, OH, OW, stride, pad;
+  FastDiv d_ohow, d_ow, d_hw, d_w, d_khkw, d_kw;  // runtime-dim magics
+};
+
+// compile-time geometry: CK=0 means runtime (generic fallback)
+template <int CKH, int CKW, int CS, int CP>
+struct Geo {
+  static constexpr bool fixed = CKH > 0;
+  __device__ static int kh_kw(const ConvGeom& g) { return fixed ? CKH * CKW : g.KH * g.KW; }
+  __device__ static int kw(const ConvGeom& g) { return fixed ? CKW : g.KW; }
+  __device__ static int stride(const ConvGeom& g) { return fixed ? CS : g.stride; }
+  __device__ static int pad(const ConvGeom& g) { return fixed ? CP : g.pad; }
 };
 
 // ---------------- forward ----------------
+template <int CKH, int CKW, int CS, int CP>
 struct ConvFwdGather {
+  using G = Geo<CKH, CKW, CS, CP>;
   const float* w;  // [Co, Ci, KH, KW]
   const float* x;  // [B, Ci, H, W]
   ConvGeom geo;
   __device__ float loadA(int, int m, int k) const {  // A[Co][Ci*KH*KW]
-    return w[(long)m * (geo.Ci * geo.KH * geo.KW) + k];
+    return w[(long)m * (geo.Ci * G::kh_kw(geo)) + k];
   }
   __device__ float loadB(int, int k, int n) const {  // B[Ci*KH*KW][B*OH*OW]
-    const int ohow = geo.OH * geo.OW;
-    const int b = n / ohow;
-    const int rem = n - b * ohow;
-    const int oh = rem / geo.OW;
-    const int ow = rem - oh * geo.OW;
-    const int ci = k / (geo.KH * geo.KW);
-    const int r = k - ci * (geo.KH * geo.KW);
-    const int kh = r / geo.KW;
-    const int kw = r - kh * geo.KW;
-    const int ih = oh * geo.stride - geo.pad + kh;
-    const int iw = ow * geo.stride - geo.pad + kw;
+    const unsigned b = geo.d_ohow.div(n);
+    const unsigned rem = geo.d_ohow.mod(n, b);
+    const unsigned oh = geo.d_ow.div(rem);
+    const unsigned ow = geo.d_ow.mod(rem, oh);
+    int ci, kh, kw;
+    if (G::fixed) {
+      ci = k / (CKH * CKW);
+      const int r = k - ci * (CKH * CKW);
+      kh = r / CKW;
+      kw = r - kh * CKW;
+    } else {
+      ci = geo.d_khkw.div(k);
+      const int r = geo.d_khkw.mod(k, ci);
+      kh = geo.d_kw.div(r);
+      kw = geo.d_kw.mod(r, kh);
+    }
+    const int ih = (int)oh * G::stride(geo) - G::pad(geo) + kh;
+    const int iw = (int)ow * G::stride(geo) - G::pad(geo) + kw;
     if (ih < 0 || ih >= geo.H || iw < 0 || iw >= geo.W) return 0.f;
     return x[(((long)b * geo.Ci + ci) * geo.H + ih) * geo.W + iw];
   }
@@ -49,43 +76,70 @@ struct ConvFwdStore {
   float* y;  // [B, Co, OH, OW]
   const float* bias;  // [Co] nullable
   int Co, OHOW;
-  __device__ void store(int, int m, int n, float v) const {
-    const int b = n / OHOW;
-    const int rem = n - b * OHOW;
-    if (bias != nullptr) v += bias[m];
-    y[((long)b * Co + m) * OHOW + rem] = v;
+  FastDiv d_ohow;
+  bool accumulate;
+  __device__ void store(int, int m, int n, float v, bool first_split) const {
+    const unsigned b = d_ohow.div(n);
+    const unsigned rem = d_ohow.mod(n, b);
+    if (bias != nullptr && first_split) v += bias[m];
+    float* p = y + ((long)b * Co + m) * OHOW + rem;
+    if (accumulate) {
+      atomicAdd(p, v);
+    } else {
+      *p = v;
+    }
   }
 };
 
 // ---------------- backward data ----------------
+template <int CKH, int CKW, int CS, int CP>
 struct ConvBwdDataGather {
+  using G = Geo<CKH, CKW, CS, CP>;
   const float* w;   // [Co, Ci, KH, KW]
   const float* gy;  // [B, Co, OH, OW]
   ConvGeom geo;
   __device__ float loadA(int, int m, int k) const {  // A[Ci][Co*KH*KW]
-    const int khkw = geo.KH * geo.KW;
-    const int co = k / khkw;
-    const int r = k - co * khkw;
+    const int khkw = G::kh_kw(geo);
+    int co, r;
+    if (G::fixed) {
+      co = k / (CKH * CKW);
+      r = k - co * (CKH * CKW);
+    } else {
+      co = geo.d_khkw.div(k);
+      r = geo.d_khkw.mod(k, co);
+    }
     return w[((long)co * geo.Ci + m) * khkw + r];
   }
   __device__ float loadB(int, int k, int n) const {  // B[Co*KH*KW][B*H*W]
-    const int hw = geo.H * geo.W;
-    const int b = n / hw;
-    const int rem = n - b * hw;
-    const int ih = rem / geo.W;
-    const int iw = rem - ih * geo.W;
-    const int khkw = geo.KH * geo.KW;
-    const int co = k / khkw;
-    const int r = k - co * khkw;
-    const int kh = r / geo.KW;
-    const int kw = r - kh * geo.KW;
-    const int oh_num = ih + geo.pad - kh;
-    const int ow_num = iw + geo.pad - kw;
-    if (oh_num < 0 || ow_num < 0) return 0.f;
-    if (oh_num % geo.stride != 0 || ow_num % geo.stride != 0) return 0.f;
-    const int oh = oh_num / geo.stride;
-    const int ow = ow_num / geo.stride;
-    if (oh >= geo.OH || ow >= geo.OW) return 0.f;
+    const unsigned b = geo.d_hw.div(n);
+    const unsigned rem = geo.d_hw.mod(n, b);
+    const unsigned ih = geo.d_w.div(rem);
+    const unsigned iw = geo.d_w.mod(rem, ih);
+    int co, kh, kw;
+    if (G::fixed) {
+      co = k / (CKH * CKW);
+      const int r = k - co * (CKH * CKW);
+      kh = r / CKW;
+      kw = r - kh * CKW;
+    } else {
+      co = geo.d_khkw.div(k);
+      const int r = geo.d_khkw.mod(k, co);
+      kh = geo.d_kw.div(r);
+      kw = geo.d_kw.mod(r, kh);
+    }
+    const int s = G::stride(geo);
+    const int oh_num = (int)ih + G::pad(geo) - kh;
+    const int ow_num = (int)iw + G::pad(geo) - kw;
+    int oh, ow;
+    if (s == 1) {
+      oh = oh_num;
+      ow = ow_num;
+    } else {
+      if ((oh_num % s) != 0 || (ow_num % s) != 0) return 0.f;
+      oh = oh_num / s;
+      ow = ow_num / s;
+    }
+    if (oh < 0 || ow < 0 || oh >= geo.OH || ow >= geo.OW) return 0.f;
     return gy[(((long)b * geo.Co + co) * geo.OH + oh) * geo.OW + ow];
   }
 };
@@ -93,37 +147,51 @@ struct ConvBwdDataGather {
 struct ConvBwdDataStore {
   float* gx;  // [B, Ci, H, W]
   int Ci, HW;
-  __device__ void store(int, int m, int n, float v) const {
-    const int b = n / HW;
-    const int rem = n - b * HW;
-    gx[((long)b * Ci + m) * HW + rem] = v;
+  FastDiv d_hw;
+  bool accumulate;
+  __device__ void store(int, int m, int n, float v, bool) const {
+    const unsigned b = d_hw.div(n);
+    const unsigned rem = d_hw.mod(n, b);
+    float* p = gx + ((long)b * Ci + m) * HW + rem;
+    if (accumulate) {
+      atomicAdd(p, v);
+    } else {
+      *p = v;
+    }
   }
 };
 
 // ---------------- backward weight ----------------
+template <int CKH, int CKW, int CS, int CP>
 struct ConvBwdWeightGather {
+  using G = Geo<CKH, CKW, CS, CP>;
   const float* gy;  // [B, Co, OH, OW]
   const float* x;   // [B, Ci, H, W]
   ConvGeom geo;
   __device__ float loadA(int, int m, int k) const {  // A[Co][B*OH*OW]
-    const int ohow = geo.OH * geo.OW;
-    const int b = k / ohow;
-    const int rem = k - b * ohow;
-    return gy[((long)b * geo.Co + m) * ohow + rem];
+    const unsigned b = geo.d_ohow.div(k);
+    const unsigned rem = geo.d_ohow.mod(k, b);
+    return gy[((long)b * geo.Co + m) * (geo.OH * geo.OW) + rem];
   }
   __device__ float loadB(int, int k, int n) const {  // B[B*OH*OW][Ci*KH*KW]
-    const int ohow = geo.OH * geo.OW;
-    const int b = k / ohow;
-    const int rem = k - b * ohow;
-    const int oh = rem / geo.OW;
-    const int ow = rem - oh * geo.OW;
-    const int khkw = geo.KH * geo.KW;
-    const int ci = n / khkw;
-    const int r = n - ci * khkw;
-    const int kh = r / geo.KW;
-    const int kw = r - kh * geo.KW;
-    const int ih = oh * geo.stride - geo.pad + kh;
-    const int iw = ow * geo.stride - geo.pad + kw;
+    const unsigned b = geo.d_ohow.div(k);
+    const unsigned rem = geo.d_ohow.mod(k, b);
+    const unsigned oh = geo.d_ow.div(rem);
+    const unsigned ow = geo.d_ow.mod(rem, oh);
+    int ci, kh, kw;
+    if (G::fixed) {
+      ci = n / (CKH * CKW);
+      const int r = n - ci * (CKH * CKW);
+      kh = r / CKW;
+      kw = r - kh * CKW;
+    } else {
+      ci = geo.d_khkw.div(n);
+      const int r = geo.d_khkw.mod(n, ci);
+      kh = geo.d_kw.div(r);
+      kw = geo.d_kw.mod(r, kh);
+    }
+    const int ih = (int)oh * G::stride(geo) - G::pad(geo) + kh;
+    const int iw = (int)ow * G::stride(geo) - G::pad(geo) + kw;
     if (ih < 0 || ih >= geo.H || iw < 0 || iw >= geo.W) return 0.f;
     return x[(((long)b * geo.Ci + ci) * geo.H + ih) * geo.W + iw];
   }
@@ -132,22 +200,57 @@ struct ConvBwdWeightGather {
 struct AtomicStore {
   float* c;  // [M, N] contiguous, pre-zeroed
   int N;
-  __device__ void store(int, int m, int n, float v) const {
+  __device__ void store(int, int m, int n, float v, bool) const {
     atomicAdd(c + (long)m * N + n, v);
   }
 };
 
 // ---------------- host wrappers ----------------
 
-static ConvGeom make_geom(const at::Tensor& x, const at::Tensor& w, int stride, int pad) {
+static ConvGeom make_geom(int B, int Ci, int H, int W, int Co, int KH, int KW,
+                          int stride, int pad) {
   ConvGeom g;
-  g.B = x.size(0); g.Ci = x.size(1); g.H = x.size(2); g.W = x.size(3);
-  g.Co = w.size(0); g.KH = w.size(2); g.KW = w.size(3);
+  g.B = B; g.Ci = Ci; g.H = H; g.W = W; g.Co = Co; g.KH = KH; g.KW = KW;
   g.stride = stride; g.pad = pad;
-  g.OH = (g.H + 2 * pad - g.KH) / stride + 1;
-  g.OW = (g.W + 2 * pad - g.KW) / stride + 1;
+  g.OH = (H + 2 * pad - KH) / stride + 1;
+  g.OW = (W + 2 * pad - KW) / stride + 1;
+  g.d_ohow.init(g.OH * g.OW);
+  g.d_ow.init(g.OW);
+  g.d_hw.init(g.H * g.W);
+  g.d_w.init(g.W);
+  g.d_khkw.init(g.KH * g.KW);
+  g.d_kw.init(g.KW);
   return g;
 }
+
+// dispatch over the model zoo's conv geometries
+template <typename F>
+static void dispatch_geom(const ConvGeom& g, F&& f) {
+  if (g.KH == 3 && g.KW == 3 && g.stride == 1 && g.pad == 1) {
+    f(std::integral_constant<int, 0>{});  // 3x3 s1 p1
+  } else if (g.KH == 3 && g.KW == 3 && g.stride == 2 && g.pad == 1) {
+    f(std::integral_constant<int, 1>{});  // 3x3 s2 p1
+  } else if (g.KH == 1 && g.KW == 1 && g.stride == 1 && g.pad == 0) {
+    f(std::integral_constant<int, 2>{});  // 1x1
+  } else if (g.KH == 4 && g.KW == 4 && g.stride == 4 && g.pad == 0) {
+    f(std::integral_constant<int, 3>{});  // ViT patch embed
+  } else {
+    f(std::integral_constant<int, 4>{});  // generic runtime geometry
+  }
+}
+
+template <template <int, int, int, int> class Gather, int CASE>
+struct PickGather;
+template <template <int, int, int, int> class Gather>
+struct PickGather<Gather, 0> { using type = Gather<3, 3, 1, 1>; };
+template <template <int, int, int, int> class Gather>
+struct PickGather<Gather, 1> { using type = Gather<3, 3, 2, 1>; };
+template <template <int, int, int, int> class Gather>
+struct PickGather<Gather, 2> { using type = Gather<1, 1, 1, 0>; };
+template <template <int, int, int, int> class Gather>
+struct PickGather<Gather, 3> { using type = Gather<4, 4, 4, 0>; };
+template <template <int, int, int, int> class Gather>
+struct PickGather<Gather, 4> { using type = Gather<0, 0, 0, 0>; };
 
 at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                       c10::optional<at::Tensor> bias, int stride, int pad) {
@@ -155,16 +258,23 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   TORCH_CHECK(w.is_cuda() && w.dim() == 4 && w.size(1) == x.size(1));
   auto xc = x.contiguous();
   auto wc = w.contiguous();
-  ConvGeom geo = make_geom(xc, wc, stride, pad);
-  auto y = at::empty({geo.B, geo.Co, geo.OH, geo.OW}, x.options());
+  ConvGeom geo = make_geom(x.size(0), x.size(1), x.size(2), x.size(3),
+                           w.size(0), w.size(2), w.size(3), stride, pad);
+  const int M = geo.Co, N = geo.B * geo.OH * geo.OW, K = geo.Ci * geo.KH * geo.KW;
+  const int split_k = slk_pick_split_k(M, N, K, 1);
+  auto y = split_k > 1
+      ? at::zeros({geo.B, geo.Co, geo.OH, geo.OW}, x.options())
+      : at::empty({geo.B, geo.Co, geo.OH, geo.OW}, x.options());
 
-  ConvFwdGather g{wc.data_ptr<float>(), xc.data_ptr<float>(), geo};
   ConvFwdStore st{y.data_ptr<float>(),
                   bias.has_value() ? bias->data_ptr<float>() : nullptr,
-                  geo.Co, geo.OH * geo.OW};
-  const int M = geo.Co, N = geo.B * geo.OH * geo.OW, K = geo.Ci * geo.KH * geo.KW;
+                  geo.Co, geo.OH * geo.OW, geo.d_ohow, split_k > 1};
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  slk_launch_gemm(g, st, M, N, K, 1, 1, stream);
+  dispatch_geom(geo, [&](auto ic) {
+    using GT = typename PickGather<ConvFwdGather, decltype(ic)::value>::type;
+    GT g{wc.data_ptr<float>(), xc.data_ptr<float>(), geo};
+    slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
+  });
   return y;
 }
 
@@ -173,17 +283,21 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
   TORCH_CHECK(gy.is_cuda() && gy.dim() == 4 && gy.scalar_type() == at::kFloat);
   auto gyc = gy.contiguous();
   auto wc = w.contiguous();
-  ConvGeom geo;
-  geo.B = gy.size(0); geo.Co = gy.size(1); geo.OH = gy.size(2); geo.OW = gy.size(3);
-  geo.Ci = w.size(1); geo.KH = w.size(2); geo.KW = w.size(3);
-  geo.H = H; geo.W = W; geo.stride = stride; geo.pad = pad;
-  auto gx = at::empty({geo.B, geo.Ci, geo.H, geo.W}, gy.options());
-
-  ConvBwdDataGather g{wc.data_ptr<float>(), gyc.data_ptr<float>(), geo};
-  ConvBwdDataStore st{gx.data_ptr<float>(), geo.Ci, geo.H * geo.W};
+  ConvGeom geo = make_geom(gy.size(0), w.size(1), H, W, gy.size(1), w.size(2),
+                           w.size(3), stride, pad);
   const int M = geo.Ci, N = geo.B * geo.H * geo.W, K = geo.Co * geo.KH * geo.KW;
+  const int split_k = slk_pick_split_k(M, N, K, 1);
+  auto gx = split_k > 1 ? at::zeros({geo.B, geo.Ci, geo.H, geo.W}, gy.options())
+                        : at::empty({geo.B, geo.Ci, geo.H, geo.W}, gy.options());
+
+  ConvBwdDataStore st{gx.data_ptr<float>(), geo.Ci, geo.H * geo.W, geo.d_hw,
+                      split_k > 1};
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  slk_launch_gemm(g, st, M, N, K, 1, 1, stream);
+  dispatch_geom(geo, [&](auto ic) {
+    using GT = typename PickGather<ConvBwdDataGather, decltype(ic)::value>::type;
+    GT g{wc.data_ptr<float>(), gyc.data_ptr<float>(), geo};
+    slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
+  });
   return gx;
 }
 
@@ -192,19 +306,19 @@ at::Tensor conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x, int KH,
   TORCH_CHECK(gy.is_cuda() && x.is_cuda() && gy.scalar_type() == at::kFloat);
   auto gyc = gy.contiguous();
   auto xc = x.contiguous();
-  ConvGeom geo;
-  geo.B = x.size(0); geo.Ci = x.size(1); geo.H = x.size(2); geo.W = x.size(3);
-  geo.Co = gy.size(1); geo.OH = gy.size(2); geo.OW = gy.size(3);
-  geo.KH = KH; geo.KW = KW; geo.stride = stride; geo.pad = pad;
-
+  ConvGeom geo = make_geom(x.size(0), x.size(1), x.size(2), x.size(3), gy.size(1),
+                           KH, KW, stride, pad);
   const int M = geo.Co, N = geo.Ci * KH * KW, K = geo.B * geo.OH * geo.OW;
   auto gw = at::zeros({geo.Co, geo.Ci, KH, KW}, gy.options());
 
-  ConvBwdWeightGather g{gyc.data_ptr<float>(), xc.data_ptr<float>(), geo};
   AtomicStore st{gw.data_ptr<float>(), N};
   int split_k = slk_pick_split_k(M, N, K, 1);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
+  dispatch_geom(geo, [&](auto ic) {
+    using GT = typename PickGather<ConvBwdWeightGather, decltype(ic)::value>::type;
+    GT g{gyc.data_ptr<float>(), xc.data_ptr<float>(), geo};
+    slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
+  });
   return gw;
 }
 

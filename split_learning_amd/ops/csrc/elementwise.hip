@@ -144,6 +144,23 @@ __global__ void dropout_fwd_kernel(const float* __restrict__ x, float* __restric
   }
 }
 
+// variant reading the offset from device memory: hipGraph-replayable (a
+// captured host constant would freeze the mask; a device counter bumped by a
+// captured add keeps masks advancing across replays)
+__global__ void dropout_fwd_dev_kernel(const float* __restrict__ x,
+                                       float* __restrict__ y,
+                                       uint8_t* __restrict__ mask, long n, float p,
+                                       float scale, uint64_t seed,
+                                       const int64_t* __restrict__ offset_ptr) {
+  const uint64_t offset = (uint64_t)(*offset_ptr);
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    const bool keep = slk_uniform(seed, offset, (uint64_t)i) >= p;
+    mask[i] = keep;
+    y[i] = keep ? x[i] * scale : 0.f;
+  }
+}
+
 __global__ void dropout_bwd_kernel(const float* __restrict__ gy,
                                    const uint8_t* __restrict__ mask,
                                    float* __restrict__ gx, long n, float scale) {
@@ -163,6 +180,20 @@ std::vector<at::Tensor> dropout_fwd(const at::Tensor& x, double p, int64_t seed,
                      stream, x.data_ptr<float>(), y.data_ptr<float>(),
                      mask.data_ptr<uint8_t>(), n, (float)p, scale, (uint64_t)seed,
                      (uint64_t)offset);
+  return {y, mask};
+}
+
+std::vector<at::Tensor> dropout_fwd_dev(const at::Tensor& x, double p, int64_t seed,
+                                        const at::Tensor& offset) {
+  auto y = at::empty_like(x);
+  auto mask = at::empty(x.sizes(), x.options().dtype(at::kByte));
+  const long n = x.numel();
+  const float scale = 1.0f / (1.0f - (float)p);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(dropout_fwd_dev_kernel, dim3(ew_grid(n, 256, 1)), dim3(256), 0,
+                     stream, x.data_ptr<float>(), y.data_ptr<float>(),
+                     mask.data_ptr<uint8_t>(), n, (float)p, scale, (uint64_t)seed,
+                     offset.data_ptr<int64_t>());
   return {y, mask};
 }
 

@@ -25,7 +25,8 @@ constexpr int SLK_LDS_PAD = 4;
 //   __device__ float loadA(int batch, int m, int k) const;  // 0 if OOB
 //   __device__ float loadB(int batch, int k, int n) const;
 // The store functor provides:
-//   __device__ void store(int batch, int m, int n, float v) const; // handles OOB
+//   __device__ void store(int batch, int m, int n, float v, bool first_split)
+//     const;  // handles OOB; first_split gates one-time epilogue work (bias)
 template <typename Gather, typename Store>
 __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     Gather g, Store st, int M, int N, int K, int split_k, int k_per_split) {
@@ -98,7 +99,7 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
       for (int i = 0; i < 4; ++i) {
         int row = m0 + wm + mi * 16 + frag_r * 4 + i;
         int col = n0 + wn + ni * 16 + frag_c;
-        if (row < M && col < N) st.store(batch, row, col, acc[mi][ni][i]);
+        if (row < M && col < N) st.store(batch, row, col, acc[mi][ni][i], ks == 0);
       }
     }
   }

@@ -227,9 +227,9 @@ def maxpool2x2(x):
 
 class DropoutFn(Function):
     @staticmethod
-    def forward(ctx, x, p, seed, offset):
+    def forward(ctx, x, p, seed, offset_t):
         x = x.contiguous()
-        y, mask = native().dropout_fwd(x, float(p), int(seed), int(offset))
+        y, mask = native().dropout_fwd_dev(x, float(p), int(seed), offset_t)
         ctx.save_for_backward(mask)
         ctx.p = float(p)
         return y
@@ -240,19 +240,31 @@ class DropoutFn(Function):
         return native().dropout_bwd(gy.contiguous(), mask, ctx.p), None, None, None
 
 
-_DROPOUT_STATE = {"seed": 0x5EEDC0DE, "offset": 0}
+# The philox-style offset lives in DEVICE memory and is bumped by a captured
+# add, so dropout masks keep advancing under hipGraph replay of the step.
+_DROPOUT_STATE = {"seed": 0x5EEDC0DE, "counters": {}}
 
 
 def seed_dropout(seed: int) -> None:
     _DROPOUT_STATE["seed"] = int(seed) & 0xFFFFFFFFFFFFFFFF
-    _DROPOUT_STATE["offset"] = 0
+    for c in _DROPOUT_STATE["counters"].values():
+        c.zero_()
+
+
+def _counter_for(device) -> torch.Tensor:
+    c = _DROPOUT_STATE["counters"].get(device)
+    if c is None:
+        c = torch.zeros(1, dtype=torch.int64, device=device)
+        _DROPOUT_STATE["counters"][device] = c
+    return c
 
 
 def dropout(x, p, training=True):
     if not training or p == 0.0:
         return x
-    _DROPOUT_STATE["offset"] += 1
-    return DropoutFn.apply(x, p, _DROPOUT_STATE["seed"], _DROPOUT_STATE["offset"])
+    counter = _counter_for(x.device)
+    counter.add_(1)
+    return DropoutFn.apply(x, p, _DROPOUT_STATE["seed"], counter)
 
 
 # ---------------------------------------------------------------------------
