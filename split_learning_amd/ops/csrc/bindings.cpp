@@ -14,7 +14,9 @@ at::Tensor conv2d_bwd_data(const at::Tensor&, const at::Tensor&, int, int, int, 
 at::Tensor conv2d_bwd_weight(const at::Tensor&, const at::Tensor&, int, int, int, int);
 at::Tensor conv2d_bwd_bias(const at::Tensor&);
 // norm.hip
-std::vector<at::Tensor> bn2d_stats(const at::Tensor&);
+std::vector<at::Tensor> bn2d_stats_fused(const at::Tensor&,
+                                         c10::optional<at::Tensor>,
+                                         c10::optional<at::Tensor>, double, double);
 at::Tensor bn2d_fwd(const at::Tensor&, const at::Tensor&, const at::Tensor&,
                     const at::Tensor&, const at::Tensor&, bool);
 std::vector<at::Tensor> bn2d_bwd(const at::Tensor&, const at::Tensor&,
@@ -50,10 +52,10 @@ std::vector<at::Tensor> ce_fwd(const at::Tensor&, const at::Tensor&);
 at::Tensor ce_bwd(const at::Tensor&, const at::Tensor&, const at::Tensor&);
 // optim.hip
 void sgd_step(std::vector<at::Tensor>, std::vector<at::Tensor>,
-              std::vector<at::Tensor>, double, double, double, bool);
+              std::vector<at::Tensor>, double, double, double, bool, bool);
 void adamw_step(std::vector<at::Tensor>, std::vector<at::Tensor>,
                 std::vector<at::Tensor>, std::vector<at::Tensor>, int64_t, double,
-                double, double, double, double);
+                double, double, double, double, bool);
 }  // namespace slk
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -67,7 +69,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_data", &slk::conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &slk::conv2d_bwd_weight);
   m.def("conv2d_bwd_bias", &slk::conv2d_bwd_bias);
-  m.def("bn2d_stats", &slk::bn2d_stats);
+  m.def("bn2d_stats_fused", &slk::bn2d_stats_fused);
   m.def("bn2d_fwd", &slk::bn2d_fwd);
   m.def("bn2d_bwd", &slk::bn2d_bwd);
   m.def("bn2d_bwd_eval", &slk::bn2d_bwd_eval);

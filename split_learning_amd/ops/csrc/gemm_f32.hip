@@ -102,11 +102,15 @@ at::Tensor matmul_f32(const at::Tensor& a, const at::Tensor& b, bool ta, bool tb
 
 at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
                       c10::optional<at::Tensor> bias) {
-  // y[M,N] = x[M,K] @ w[N,K]^T + b
+  // y[M,N] = x[M,K] @ w[N,K]^T + b; split-K when the tile grid underfills the
+  // chip (batch-32 rows = 1 M-tile: the 4096->10 classifier is ONE tile).
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat, "linear_fwd: fp32 GPU");
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1));
-  auto y = at::empty({x.size(0), w.size(0)}, x.options());
-  launch_strided(x, w, y, /*ta=*/false, /*tb=*/true, bias, /*acc=*/false, 1);
+  const int M = x.size(0), N = w.size(0), K = x.size(1);
+  int split_k = slk_pick_split_k(M, N, K, 1);
+  auto y = split_k > 1 ? at::zeros({M, N}, x.options())
+                       : at::empty({M, N}, x.options());
+  launch_strided(x, w, y, /*ta=*/false, /*tb=*/true, bias, /*acc=*/false, split_k);
   return y;
 }
 

@@ -13,13 +13,21 @@ namespace slk {
 
 // ---------------- BatchNorm2d ----------------
 
-// one block per channel; double accumulators for stable mean/var at any N*H*W
-__global__ void bn_stats_kernel(const float* __restrict__ x, float* __restrict__ mean,
-                                float* __restrict__ var, int B, int C, int HW) {
+// chunked partial sums (grid.y chunks per channel -> float atomics) followed
+// by a finalize kernel that also folds invstd and the running-stat update
+// in-kernel (the previous host-side rsqrt/mul_/add_ chain was 5 extra
+// kernel launches per BN call).
+__global__ void bn_partial_kernel(const float* __restrict__ x,
+                                  float* __restrict__ sum, float* __restrict__ sumsq,
+                                  int B, int C, int HW) {
   __shared__ double scratch[16];
   const int c = blockIdx.x;
+  const int total = B * HW;
+  const int per = (total + gridDim.y - 1) / gridDim.y;
+  const int lo = blockIdx.y * per;
+  const int hi = min(total, lo + per);
   double s = 0.0, s2 = 0.0;
-  for (int i = threadIdx.x; i < B * HW; i += blockDim.x) {
+  for (int i = lo + threadIdx.x; i < hi; i += blockDim.x) {
     const int b = i / HW;
     const int r = i - b * HW;
     const double v = (double)x[((long)b * C + c) * HW + r];
@@ -30,10 +38,28 @@ __global__ void bn_stats_kernel(const float* __restrict__ x, float* __restrict__
   __syncthreads();
   double ts2 = slk_block_sum(s2, scratch);
   if (threadIdx.x == 0) {
-    const double n = (double)B * HW;
-    const double m = ts / n;
-    mean[c] = (float)m;
-    var[c] = (float)fmax(ts2 / n - m * m, 0.0);
+    atomicAdd(sum + c, (float)ts);
+    atomicAdd(sumsq + c, (float)ts2);
+  }
+}
+
+__global__ void bn_finalize_kernel(const float* __restrict__ sum,
+                                   const float* __restrict__ sumsq,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ invstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var, int C, float n,
+                                   float momentum, float eps) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float m = sum[c] / n;
+  const float v = fmaxf(sumsq[c] / n - m * m, 0.f);
+  mean[c] = m;
+  invstd[c] = rsqrtf(v + eps);
+  if (running_mean != nullptr) {
+    const float unbiased = v * (n / fmaxf(n - 1.f, 1.f));
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
   }
 }
 
@@ -104,15 +130,35 @@ __global__ void bn_bwd_dx_kernel(const float* __restrict__ x,
   }
 }
 
-std::vector<at::Tensor> bn2d_stats(const at::Tensor& x) {
+static inline int bn_chunks(long per_channel) {
+  long c = per_channel / 4096;
+  if (c < 1) c = 1;
+  if (c > 16) c = 16;
+  return (int)c;
+}
+
+std::vector<at::Tensor> bn2d_stats_fused(const at::Tensor& x,
+                                         c10::optional<at::Tensor> running_mean,
+                                         c10::optional<at::Tensor> running_var,
+                                         double momentum, double eps) {
   const int B = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto mean = at::empty({C}, x.options());
-  auto var = at::empty({C}, x.options());
+  auto invstd = at::empty({C}, x.options());
+  auto sums = at::zeros({2, C}, x.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(bn_stats_kernel, dim3(C), dim3(256), 0, stream,
-                     x.data_ptr<float>(), mean.data_ptr<float>(),
-                     var.data_ptr<float>(), B, C, HW);
-  return {mean, var};
+  const int chunks = bn_chunks((long)B * HW);
+  hipLaunchKernelGGL(bn_partial_kernel, dim3(C, chunks), dim3(256), 0, stream,
+                     x.data_ptr<float>(), sums.data_ptr<float>(),
+                     sums.data_ptr<float>() + C, B, C, HW);
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
+                     stream, sums.data_ptr<float>(), sums.data_ptr<float>() + C,
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     running_mean.has_value() ? running_mean->data_ptr<float>()
+                                              : nullptr,
+                     running_var.has_value() ? running_var->data_ptr<float>()
+                                             : nullptr,
+                     C, (float)((long)B * HW), (float)momentum, (float)eps);
+  return {mean, invstd};
 }
 
 at::Tensor bn2d_fwd(const at::Tensor& x, const at::Tensor& mean,

@@ -11,12 +11,16 @@
 
 namespace slk {
 
-__global__ void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
+// zero_after: clears the gradient after applying it, replacing the per-tensor
+// zero_grad fill kernels (autograd then accumulates the next microbatch's
+// gradients into already-zeroed buffers).
+__global__ void sgd_kernel(float* __restrict__ p, float* __restrict__ g,
                            float* __restrict__ buf, long n, float lr, float momentum,
-                           float weight_decay, bool first) {
+                           float weight_decay, bool first, bool zero_after) {
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     float grad = g[i];
+    if (zero_after) g[i] = 0.f;
     if (weight_decay != 0.f) grad += weight_decay * p[i];
     float b;
     if (momentum != 0.f) {
@@ -29,15 +33,17 @@ __global__ void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
   }
 }
 
-__global__ void adamw_kernel(float* __restrict__ p, const float* __restrict__ g,
+__global__ void adamw_kernel(float* __restrict__ p, float* __restrict__ g,
                              float* __restrict__ m, float* __restrict__ v, long n,
                              float lr, float beta1, float beta2, float eps,
-                             float weight_decay, float bc1, float bc2) {
+                             float weight_decay, float bc1, float bc2,
+                             bool zero_after) {
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     // decoupled weight decay (AdamW): p *= (1 - lr*wd)
     float pv = p[i] * (1.f - lr * weight_decay);
     const float grad = g[i];
+    if (zero_after) g[i] = 0.f;
     const float mi = beta1 * m[i] + (1.f - beta1) * grad;
     const float vi = beta2 * v[i] + (1.f - beta2) * grad * grad;
     m[i] = mi;
@@ -54,21 +60,21 @@ static inline int opt_grid(long n) {
 
 void sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
               std::vector<at::Tensor> bufs, double lr, double momentum,
-              double weight_decay, bool first) {
+              double weight_decay, bool first, bool zero_after) {
   auto stream = c10::hip::getCurrentHIPStream().stream();
   for (size_t i = 0; i < params.size(); ++i) {
     const long n = params[i].numel();
     hipLaunchKernelGGL(sgd_kernel, dim3(opt_grid(n)), dim3(256), 0, stream,
                        params[i].data_ptr<float>(), grads[i].data_ptr<float>(),
                        bufs[i].data_ptr<float>(), n, (float)lr, (float)momentum,
-                       (float)weight_decay, first);
+                       (float)weight_decay, first, zero_after);
   }
 }
 
 void adamw_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                 std::vector<at::Tensor> ms, std::vector<at::Tensor> vs, int64_t step,
                 double lr, double beta1, double beta2, double eps,
-                double weight_decay) {
+                double weight_decay, bool zero_after) {
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const float bc1 = 1.f - powf((float)beta1, (float)step);
   const float bc2 = 1.f - powf((float)beta2, (float)step);
@@ -78,7 +84,7 @@ void adamw_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                        params[i].data_ptr<float>(), grads[i].data_ptr<float>(),
                        ms[i].data_ptr<float>(), vs[i].data_ptr<float>(), n, (float)lr,
                        (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
-                       bc1, bc2);
+                       bc1, bc2, zero_after);
   }
 }
 

@@ -119,12 +119,14 @@ inline void slk_launch_gemm(const Gather& g, const Store& st, int M, int N, int 
                      stream, g, st, M, N, K, split_k, k_per_split);
 }
 
-// Heuristic: pick split_k so the grid roughly fills 256 CUs (8 XCDs).
+// Heuristic: pick split_k so the grid oversubscribes the 256 CUs (~3 blocks/CU
+// target: at 1 block/CU only 4 waves are resident and the gather loads stall
+// the MFMAs — measured in profiles/).
 inline int slk_pick_split_k(int M, int N, int K, int n_batch) {
   long tiles = (long)ceil_div(M, SLK_BM) * ceil_div(N, SLK_BN) * (n_batch > 0 ? n_batch : 1);
-  if (tiles >= 256 || K <= SLK_BK * 2) return 1;
-  long want = 256 / tiles;
-  long maxk = (K + 4 * SLK_BK - 1) / (4 * SLK_BK);  // keep >=4 BK steps per split
+  if (tiles >= 768 || K <= SLK_BK * 2) return 1;
+  long want = (768 + tiles - 1) / tiles;
+  long maxk = (K + 2 * SLK_BK - 1) / (2 * SLK_BK);  // keep >=2 BK steps per split
   long sk = want < maxk ? want : maxk;
   return (int)(sk < 1 ? 1 : sk);
 }

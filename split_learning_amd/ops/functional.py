@@ -105,14 +105,10 @@ class BatchNorm2dFn(Function):
                 fuse_relu):
         x = x.contiguous()
         if training:
-            mean, var = native().bn2d_stats(x)  # biased var over (N,H,W)
-            invstd = (var + eps).rsqrt()
-            # running-stat update matches torch: unbiased var for running_var
-            with torch.no_grad():
-                n = x.numel() / x.shape[1]
-                unbiased = var * (n / max(n - 1.0, 1.0))
-                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
-                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+            # fused kernel: batch mean + biased-var invstd + torch-exact
+            # running-stat update (unbiased var), all device-side
+            mean, invstd = native().bn2d_stats_fused(x, running_mean, running_var,
+                                                     momentum, eps)
         else:
             mean = running_mean
             invstd = (running_var + eps).rsqrt()
@@ -370,16 +366,18 @@ def embedding(ids, weight, padding_idx=None):
 # ---------------------------------------------------------------------------
 
 def sgd_step(params, grads, momentum_bufs, lr, momentum, weight_decay=0.0,
-             first_step=False):
+             first_step=False, zero_grad_after=False):
     """Fused multi-tensor SGD+momentum matching torch.optim.SGD semantics
-    (buf = m*buf + g; p -= lr*buf; first step: buf = g)."""
+    (buf = m*buf + g; p -= lr*buf; first step: buf = g).  zero_grad_after
+    clears grads in the same kernel (replaces zero_grad fills)."""
     native().sgd_step(list(params), list(grads), list(momentum_bufs),
-                      float(lr), float(momentum), float(weight_decay), bool(first_step))
+                      float(lr), float(momentum), float(weight_decay),
+                      bool(first_step), bool(zero_grad_after))
 
 
 def adamw_step(params, grads, exp_avgs, exp_avg_sqs, step, lr, beta1, beta2, eps,
-               weight_decay):
+               weight_decay, zero_grad_after=False):
     """Fused multi-tensor AdamW matching torch.optim.AdamW."""
     native().adamw_step(list(params), list(grads), list(exp_avgs), list(exp_avg_sqs),
                         int(step), float(lr), float(beta1), float(beta2), float(eps),
-                        float(weight_decay))
+                        float(weight_decay), bool(zero_grad_after))

@@ -26,9 +26,9 @@ class FusedSGD:
         self._torch_opt = None
 
     def zero_grad(self):
-        for p in self.params:
-            if p.grad is not None:
-                p.grad.zero_()
+        """No-op by design: step() zeroes grads in the update kernel itself
+        (zero_grad_after), so autograd accumulates each microbatch into
+        already-zeroed buffers without extra fill launches."""
 
     @torch.no_grad()
     def step(self):
@@ -38,7 +38,8 @@ class FusedSGD:
         if live[0][0].is_cuda:
             hf.sgd_step([p for p, _ in live], [p.grad for p, _ in live],
                         [b for _, b in live], self.lr, self.momentum,
-                        self.weight_decay, first_step=(self.steps == 0))
+                        self.weight_decay, first_step=(self.steps == 0),
+                        zero_grad_after=True)
         else:
             for p, buf in live:
                 g = p.grad
@@ -51,6 +52,7 @@ class FusedSGD:
                         buf.mul_(self.momentum).add_(g)
                     g = buf
                 p.add_(g, alpha=-self.lr)
+                p.grad.zero_()
         self.steps += 1
 
 
@@ -67,9 +69,7 @@ class FusedAdamW:
         self.steps = 0
 
     def zero_grad(self):
-        for p in self.params:
-            if p.grad is not None:
-                p.grad.zero_()
+        """No-op: see FusedSGD.zero_grad."""
 
     @torch.no_grad()
     def step(self):
@@ -82,7 +82,7 @@ class FusedAdamW:
             hf.adamw_step([p for p, _, _ in live], [p.grad for p, _, _ in live],
                           [m for _, m, _ in live], [v for _, _, v in live],
                           self.steps, self.lr, self.beta1, self.beta2, self.eps,
-                          self.weight_decay)
+                          self.weight_decay, zero_grad_after=True)
         else:
             b1, b2 = self.beta1, self.beta2
             bc1 = 1 - b1 ** self.steps
@@ -92,6 +92,7 @@ class FusedAdamW:
                 m.mul_(b1).add_(p.grad, alpha=1 - b1)
                 v.mul_(b2).addcmul_(p.grad, p.grad, value=1 - b2)
                 p.addcdiv_(m / bc1, (v / bc2).sqrt().add_(self.eps), value=-self.lr)
+                p.grad.zero_()
 
 
 def make_optimizer(model_name: str, params, learning: dict):
