@@ -67,13 +67,16 @@ class ClientRuntime:
         self.n_stages = None
         self.optimizer = None
         self._routing = None
+        self.select = True           # FLEX manual select flag (client --s)
+        self.out_cluster = None      # 2LS two-level clusters
 
     # ------------------------------------------------------------------
     def register(self):
         self.control.send("server", {
             "action": "REGISTER", "client_id": self.client_id,
             "layer_id": self.layer_id, "profile": self.profile,
-            "cluster": self.cluster,
+            "cluster": self.cluster, "select": self.select,
+            "out_cluster": self.out_cluster,
         })
 
     def run(self, max_batches: Optional[int] = None, on_step=None):
@@ -132,6 +135,11 @@ class ClientRuntime:
                                             self.label_count, train=True,
                                             seed=self.client_id)
 
+        # per-round scheduler policy overrides travel in START (epochs,
+        # sync-first, sda-size, time limit — the variant policies set these)
+        overrides = msg.get("scheduler") or {}
+        self.scheduler_cfg = {**self.scheduler_cfg, **overrides}
+
         routing = msg.get("routing")
         if routing is not None and self.plane_factory is not None:
             if routing != self._routing:
@@ -150,9 +158,13 @@ class ClientRuntime:
             max_batches=max_batches,
             time_limit_s=sch.get("limited-time"),
             clip_grad_norm=sch.get("clip-grad-norm"),
+            epochs=int(sch.get("epochs", 1)),
+            sync_first=bool(sch.get("sync-first", False)),
+            sda_size=int(sch.get("sda-size", 1)),
             on_step=on_step,
         )
         result, size = run_stage(ctx)
+        pause_msg = ctx.pause_msg
 
         if self.layer_id == 1:
             # reference: first stage notifies, then blocks for PAUSE
@@ -164,17 +176,27 @@ class ClientRuntime:
             while True:
                 msg = self.inbox.recv(block=True)
                 if msg.get("action") == "PAUSE":
+                    pause_msg = msg
                     break
                 deferred.append(msg)
             for m in deferred:
                 self.inbox.push_back(m)
 
-        model = self.model
-        if self.model_name == "BERT":
-            model = merge_and_unload(model)
-            self.model = model
-        sd = copy.deepcopy(model.state_dict())
-        sd = {k: v.detach().to("cpu") for k, v in sd.items()}
+        # FLEX periodic aggregation: PAUSE may carry send=False, in which case
+        # the parameter upload is skipped this round (other/FLEX/src/Server.py:
+        # 138-143, src/RpcClient.py:103-120)
+        send_params = True
+        if pause_msg is not None and pause_msg.get("send") is False:
+            send_params = False
+
+        sd = None
+        if send_params:
+            model = self.model
+            if self.model_name == "BERT":
+                model = merge_and_unload(model)
+                self.model = model
+            sd = copy.deepcopy(model.state_dict())
+            sd = {k: v.detach().to("cpu") for k, v in sd.items()}
         self.control.send("server", {
             "action": "UPDATE", "client_id": self.client_id,
             "layer_id": self.layer_id, "cluster": self.cluster,

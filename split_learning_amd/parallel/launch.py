@@ -20,6 +20,7 @@ from ..utils import Logger
 from .client import ClientRuntime
 from .control import InProcControl
 from .data_plane import LoopbackData, P2PData
+from .policies import make_server
 from .server import Server
 
 
@@ -49,20 +50,26 @@ def assign_clients(config) -> List[dict]:
 def run_loopback(config: Dict[str, Any], device: str = "cpu",
                  max_batches: Optional[int] = None, on_step=None,
                  checkpoint_dir: str = ".", logger: Optional[Logger] = None,
-                 scheduler_cfg: Optional[dict] = None):
+                 scheduler_cfg: Optional[dict] = None,
+                 client_specs: Optional[List[dict]] = None):
     """Run the full protocol in-process. Returns (server, runtimes)."""
     control = InProcControl()
     plane = LoopbackData()
     logger = logger or Logger(f"{config['log_path']}/app.log", config["debug_mode"])
-    server = Server(config, control, logger=logger, checkpoint_dir=checkpoint_dir)
+    server = make_server(config, control, logger=logger,
+                         checkpoint_dir=checkpoint_dir)
 
     runtimes = []
     threads = []
     dev = torch.device(device)
-    for rec in assign_clients(config):
+    for rec in (client_specs or assign_clients(config)):
         rt = ClientRuntime(rec["client_id"], rec["layer_id"], control, plane,
-                           dev, cluster=rec["cluster"], logger=logger,
+                           dev, cluster=rec.get("cluster"), logger=logger,
                            scheduler_cfg=scheduler_cfg or config.get("scheduler"))
+        if rec.get("out_cluster") is not None:
+            rt.out_cluster = rec["out_cluster"]
+        if rec.get("select") is not None:
+            rt.select = rec["select"]
         runtimes.append(rt)
         t = threading.Thread(target=rt.run,
                              kwargs={"max_batches": max_batches, "on_step": on_step},
@@ -141,8 +148,8 @@ def run_p2p_client(config: Dict[str, Any], rank: int, world: int,
         # a single socket and must not be shared across threads)
         server_control = StoreControl.create(store_addr, store_port,
                                              is_server=False)
-        server = Server(config, server_control, logger=logger,
-                        checkpoint_dir=checkpoint_dir)
+        server = make_server(config, server_control, logger=logger,
+                             checkpoint_dir=checkpoint_dir)
         server_thread = threading.Thread(target=server.run, daemon=True)
         server_thread.start()
 

@@ -10,9 +10,9 @@ relay stage exists.
 
 MI355X-native differences from the reference:
 * activations/gradients stay GPU-resident end to end (no .cpu().numpy());
-* by default the stage-1 autograd graph is STASHED (288 GB HBM3E makes the
-  reference's recompute-on-backward, src/train/VGG16.py:89-91, unnecessary);
-  set scheduler.recompute for reference-exact behaviour;
+* stage-1 stashes only the INPUT batch and recomputes the forward with current
+  weights at backward (reference semantics, src/train/VGG16.py:89-91) — but the
+  initial forward runs under no_grad, which the reference wastes a graph on;
 * NaN detection accumulates in a device flag, synced once per round (the
   reference syncs every batch via loss.item(), src/train/VGG16.py:168-171).
 
@@ -63,8 +63,14 @@ class StageContext:
     max_batches: Optional[int] = None
     time_limit_s: Optional[float] = None
     clip_grad_norm: Optional[float] = None
+    epochs: int = 1                 # dataset passes per round (Vanilla_SL
+                                    # Scheduler epoch loop, capped at 100)
+    sync_first: bool = False        # DCSL/FLEX: strictly synchronous stage 1
+                                    # (block for each batch's gradient)
+    sda_size: int = 1               # DCSL SDA batching at the last stage
     on_step: Any = None             # callback(step_idx) for benchmarking hooks
     log_loss: Any = None            # callback(loss_tensor) optional
+    pause_msg: Optional[dict] = None  # PAUSE message that ended the stage loop
 
 
 def _check_pause(ctx: StageContext) -> Optional[dict]:
@@ -72,13 +78,24 @@ def _check_pause(ctx: StageContext) -> Optional[dict]:
     return msg
 
 
+def _epoch_iter(loader, epochs, max_batches):
+    """Chain `epochs` passes over the loader (reference Vanilla_SL caps the
+    epoch loop at 100, other/Vanilla_SL/src/Scheduler.py:77)."""
+    def gen():
+        for _ in range(max(1, min(int(epochs), 100))):
+            for item in loader:
+                yield item
+    it = gen()
+    if max_batches is not None:
+        it = itertools.islice(it, max_batches)
+    return it
+
+
 def train_first_stage(ctx: StageContext):
     """Returns (result, n_samples)."""
     model, opt, plane = ctx.model, ctx.optimizer, ctx.plane
-    cc = int(ctx.learning.get("control-count", 3))
-    batch_iter = iter(ctx.train_loader)
-    if ctx.max_batches is not None:
-        batch_iter = itertools.islice(batch_iter, ctx.max_batches)
+    cc = 1 if ctx.sync_first else int(ctx.learning.get("control-count", 3))
+    batch_iter = _epoch_iter(ctx.train_loader, ctx.epochs, ctx.max_batches)
     inflight: Dict[int, Any] = {}
     n_fwd = n_bwd = 0
     data_count = 0
@@ -130,37 +147,59 @@ def train_first_stage(ctx: StageContext):
 
 
 def train_last_stage(ctx: StageContext):
+    """Last stage.  sda_size > 1 enables DCSL's SDA batching
+    (other/DCSL/src/Scheduler.py:152-191): collect one activation from each of
+    sda_size upstream clients, concatenate into one super-batch, single
+    fwd/bwd, then split the cut-layer gradient back per client."""
     model, opt, plane = ctx.model, ctx.optimizer, ctx.plane
     data_count = 0
     nan_flag = torch.zeros((), dtype=torch.bool, device=ctx.device)
     model.train()
+    group: list = []
+
+    def process(batch_msgs):
+        nonlocal data_count
+        acts = [m.data.to(ctx.device, non_blocking=True) for m in batch_msgs]
+        act = (torch.cat(acts, dim=0) if len(acts) > 1 else acts[0]).detach()
+        act.requires_grad_(True)
+        labels = torch.cat([m.labels.to(ctx.device, non_blocking=True)
+                            for m in batch_msgs], dim=0)
+        opt.zero_grad()
+        out = model(act)
+        loss = _cross_entropy(out, labels)
+        nan_flag.copy_(nan_flag | torch.isnan(loss))
+        if ctx.log_loss is not None:
+            ctx.log_loss(loss)
+        loss.backward()
+        if ctx.clip_grad_norm:
+            torch.nn.utils.clip_grad_norm_(
+                [p for p in model.parameters() if p.grad is not None],
+                ctx.clip_grad_norm)
+        opt.step()
+        data_count += act.shape[0]
+        off = 0
+        for m in batch_msgs:
+            n = m.data.shape[0]
+            g = act.grad[off:off + n].detach()
+            off += n
+            plane.send_gradient(ctx.layer_id - 1, m.trace[-1],
+                                GradientMsg(m.data_id, g, m.trace[:-1]))
+
     while True:
         m = plane.recv_activation(ctx.layer_id - 1, ctx.cluster, ctx.client_id,
                                   block=False)
         if m is not None:
-            act = m.data.to(ctx.device, non_blocking=True).detach().requires_grad_(True)
-            labels = m.labels.to(ctx.device, non_blocking=True)
-            opt.zero_grad()
-            out = model(act)
-            loss = _cross_entropy(out, labels)
-            nan_flag |= torch.isnan(loss)
-            if ctx.log_loss is not None:
-                ctx.log_loss(loss)
-            loss.backward()
-            if ctx.clip_grad_norm:
-                torch.nn.utils.clip_grad_norm_(
-                    [p for p in model.parameters() if p.grad is not None],
-                    ctx.clip_grad_norm)
-            opt.step()
-            data_count += act.shape[0]
-            # queue key is the RECEIVER's stage (reference
-            # gradient_queue_{layer_id-1}_{to_client}, src/train/VGG16.py:43)
-            plane.send_gradient(
-                ctx.layer_id - 1, m.trace[-1],
-                GradientMsg(m.data_id, act.grad.detach(), m.trace[:-1]))
+            group.append(m)
+            if len(group) >= max(1, ctx.sda_size):
+                process(group)
+                group = []
         else:
             msg = _check_pause(ctx)
             if msg is not None and msg.get("action") == "PAUSE":
+                if group:  # flush a partial SDA group at round end
+                    process(group)
+                    group = []
+                ctx.pause_msg = msg
                 result = not bool(nan_flag.item())
                 return result, data_count
 
@@ -205,6 +244,7 @@ def train_middle_stage(ctx: StageContext):
         if not inflight:
             msg = _check_pause(ctx)
             if msg is not None and msg.get("action") == "PAUSE":
+                ctx.pause_msg = msg
                 return True, data_count
 
 

@@ -133,7 +133,8 @@ class Server:
     def on_register(self, msg):
         rec = {"client_id": msg["client_id"], "layer_id": msg["layer_id"],
                "profile": msg.get("profile") or {}, "cluster": msg.get("cluster"),
-               "label": [], "train": True}
+               "label": [], "train": bool(msg.get("select", True)),
+               "out_cluster": msg.get("out_cluster")}
         if self.size_data is None and rec["layer_id"] == 1:
             self.size_data = rec["profile"].get("size_data")
         if not any(c["client_id"] == rec["client_id"] for c in self.list_clients):
@@ -141,11 +142,14 @@ class Server:
             self.register_counts[rec["layer_id"] - 1] += 1
         if self.register_counts == self.total_clients:
             self._log(f"All {sum(self.total_clients)} clients registered")
-            self.distribution()
-            self.cluster_and_selection()
-            self._log(f"cut layers: {self.list_cut_layers}, clusters: {self.infor_cluster}")
-            self._log(f"Start training round {self.global_round - self.round + 1}")
-            self.notify_clients()
+            self.on_all_registered()
+
+    def on_all_registered(self):
+        self.distribution()
+        self.cluster_and_selection()
+        self._log(f"cut layers: {self.list_cut_layers}, clusters: {self.infor_cluster}")
+        self._log(f"Start training round {self.global_round - self.round + 1}")
+        self.notify_clients()
 
     def distribution(self):
         """IID or Dirichlet non-IID per-label sample counts for layer-1 clients
@@ -211,6 +215,11 @@ class Server:
                     self.list_cut_layers.append(
                         list(self.manual["no-cluster"]["cut-layers"]))
         else:
+            # manual select/reject (FLEX --s): clients that registered with
+            # select=False are rejected up front
+            for rec in self.list_clients:
+                if not rec["train"]:
+                    self.total_clients[rec["layer_id"] - 1] -= 1
             if self.manual["cluster-mode"]:
                 self.num_cluster = self.manual["cluster"]["num-cluster"]
                 self.infor_cluster = [list(r) for r in self.manual["cluster"]["infor-cluster"]]
@@ -260,49 +269,59 @@ class Server:
             "batch": batch,
         }
 
-    def notify_clients(self, start: bool = True):
-        full_state = None
-        if start and self.save_parameters and self.load_parameters and os.path.exists(self.ckpt_path):
-            full_state = torch.load(self.ckpt_path, weights_only=True)
+    # -- reusable protocol pieces (policies compose these differently) -----
+    def _load_ckpt(self):
+        if self.save_parameters and self.load_parameters and os.path.exists(self.ckpt_path):
             self._log(f"Loaded checkpoint {self.ckpt_path}")
+            return torch.load(self.ckpt_path, weights_only=True)
+        return None
 
-        active = [c for c in self.list_clients if c["train"]]
-        for rec in self.list_clients:
-            if not start:
-                self.control.send(f"client_{rec['client_id']}",
-                                  {"action": "STOP", "message": "Stop training!",
-                                   "parameters": None})
-                continue
-            if not rec["train"]:
-                if not self.reject:
-                    self.control.send(f"client_{rec['client_id']}",
-                                      {"action": "STOP", "message": "Reject Device",
-                                       "parameters": None})
-                continue
-            cuts = self.list_cut_layers[rec["cluster"]]
-            ranges = stage_ranges(cuts, self.n_stages,
-                                  get_model_class(self.model_name, self.data_name).TOTAL_UNITS)
-            layers = ranges[rec["layer_id"] - 1]
-            state = None
-            if full_state is not None:
-                part = build_partition(self.model_name, self.data_name, layers)
-                state = {k: full_state[k] for k in part.state_dict().keys()}
-            self.control.send(f"client_{rec['client_id']}", {
-                "action": "START", "message": "Server accept the connection!",
-                "parameters": state, "layers": layers,
-                "model_name": self.model_name, "data_name": self.data_name,
-                "learning": self.learning, "label_count": rec["label"],
-                "refresh": self.refresh, "cluster": rec["cluster"],
-                "n_stages": self.n_stages, "routing": self._routing_for(rec),
-            })
-        if not start:
-            self.stopped = True
-            return
-        self.reject = True
+    def _stage_layers(self, rec):
+        cuts = self.list_cut_layers[rec["cluster"]]
+        ranges = stage_ranges(cuts, self.n_stages,
+                              get_model_class(self.model_name, self.data_name).TOTAL_UNITS)
+        return ranges[rec["layer_id"] - 1]
 
-        # READY/SYN rendezvous (replaces the reference's time.sleep(25))
+    def _slice_state(self, full_state, layers):
+        if full_state is None:
+            return None
+        part = build_partition(self.model_name, self.data_name, layers)
+        return {k: full_state[k] for k in part.state_dict().keys()}
+
+    def scheduler_overrides(self, rec) -> dict:
+        """Per-policy stage-loop options shipped in START (see policies.py)."""
+        sch = self.config.get("scheduler") or {}
+        return {k: sch[k] for k in ("epochs", "limited-time", "clip-grad-norm")
+                if sch.get(k) is not None}
+
+    def _send_start(self, rec, full_state, state_override=None):
+        layers = self._stage_layers(rec)
+        state = state_override if state_override is not None             else self._slice_state(full_state, layers)
+        self.control.send(f"client_{rec['client_id']}", {
+            "action": "START", "message": "Server accept the connection!",
+            "parameters": state, "layers": layers,
+            "model_name": self.model_name, "data_name": self.data_name,
+            "learning": self.learning, "label_count": rec["label"],
+            "refresh": self.refresh, "cluster": rec["cluster"],
+            "n_stages": self.n_stages, "routing": self._routing_for(rec),
+            "scheduler": self.scheduler_overrides(rec),
+        })
+
+    def _send_stop(self, rec, message="Stop training!"):
+        self.control.send(f"client_{rec['client_id']}",
+                          {"action": "STOP", "message": message, "parameters": None})
+
+    def _send_pause(self, rec, send=True):
+        self.control.send(f"client_{rec['client_id']}",
+                          {"action": "PAUSE",
+                           "message": "Pause training and please send your parameters",
+                           "parameters": None, "send": send})
+
+    def _wait_ready(self, n):
+        """READY/SYN rendezvous (replaces the reference's time.sleep(25),
+        src/Server.py:289)."""
         ready = 0
-        while ready < len(active):
+        while ready < n:
             msg = self.control.recv("server", block=True, timeout=600.0)
             if msg is None:
                 raise TimeoutError("server: waiting for READY")
@@ -310,9 +329,30 @@ class Server:
                 ready += 1
             else:
                 self.dispatch(msg)
-        for rec in active:
+
+    def _send_syn(self, recs):
+        for rec in recs:
             self.control.send(f"client_{rec['client_id']}",
                               {"action": "SYN", "message": "Synchronize client devices"})
+
+    def notify_clients(self, start: bool = True):
+        full_state = self._load_ckpt() if start else None
+        active = [c for c in self.list_clients if c["train"]]
+        for rec in self.list_clients:
+            if not start:
+                self._send_stop(rec)
+                continue
+            if not rec["train"]:
+                if not self.reject:
+                    self._send_stop(rec, "Reject Device")
+                continue
+            self._send_start(rec, full_state)
+        if not start:
+            self.stopped = True
+            return
+        self.reject = True
+        self._wait_ready(len(active))
+        self._send_syn(active)
 
     # ------------------------------------------------------------------
     def on_notify(self, msg):
@@ -324,10 +364,7 @@ class Server:
             self._log(f"Cluster {cluster} finished; sending PAUSE")
             for rec in self.list_clients:
                 if rec["train"] and rec["cluster"] == cluster:
-                    self.control.send(f"client_{rec['client_id']}",
-                                      {"action": "PAUSE",
-                                       "message": "Pause training and please send your parameters",
-                                       "parameters": None})
+                    self._send_pause(rec)
 
     def on_update(self, msg):
         layer_id = msg["layer_id"]
