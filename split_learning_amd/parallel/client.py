@@ -69,6 +69,10 @@ class ClientRuntime:
         self._routing = None
         self.select = True           # FLEX manual select flag (client --s)
         self.out_cluster = None      # 2LS two-level clusters
+        self._fedavg_mode = "control"
+        self._group_spec = None      # cached (cluster,stage) group layout
+        self._my_group = None
+        self._my_group_ids = None
 
     # ------------------------------------------------------------------
     def register(self):
@@ -146,6 +150,26 @@ class ClientRuntime:
                 self.plane = self.plane_factory(routing)
                 self._routing = routing
 
+        # RCCL all-reduce FedAvg: build the (cluster, stage) communicators —
+        # collective, so every rank replays the same new_group sequence
+        self._fedavg_mode = msg.get("fedavg", "control")
+        groups = msg.get("fedavg_groups")
+        if self._fedavg_mode == "rccl" and groups is not None:
+            import torch.distributed as dist
+            if dist.is_available() and dist.is_initialized():
+                spec = tuple(tuple(g) for g in groups)
+                if spec != self._group_spec:
+                    self._group_spec = spec
+                    self._my_group = None
+                    self._my_group_ids = None
+                    for ids in groups:
+                        g = dist.new_group(list(ids))
+                        if self.client_id in ids:
+                            self._my_group = g
+                            self._my_group_ids = list(ids)
+            else:
+                self._fedavg_mode = "control"
+
     def _handle_syn(self, max_batches=None, on_step=None):
         sch = self.scheduler_cfg
         ctx = StageContext(
@@ -195,8 +219,19 @@ class ClientRuntime:
             if self.model_name == "BERT":
                 model = merge_and_unload(model)
                 self.model = model
-            sd = copy.deepcopy(model.state_dict())
-            sd = {k: v.detach().to("cpu") for k, v in sd.items()}
+            if self._fedavg_mode == "rccl" and self._my_group is not None:
+                # group all-reduce weighted average (xGMI); only the group
+                # representative ships the (already averaged) dict to the server
+                from .fedavg import allreduce_fedavg_
+                allreduce_fedavg_(model, float(max(size, 1)), group=self._my_group)
+                if self.client_id != min(self._my_group_ids):
+                    sd = None
+                else:
+                    sd = {k: v.detach().to("cpu")
+                          for k, v in model.state_dict().items()}
+            else:
+                sd = copy.deepcopy(model.state_dict())
+                sd = {k: v.detach().to("cpu") for k, v in sd.items()}
         self.control.send("server", {
             "action": "UPDATE", "client_id": self.client_id,
             "layer_id": self.layer_id, "cluster": self.cluster,

@@ -294,6 +294,22 @@ class Server:
         return {k: sch[k] for k in ("epochs", "limited-time", "clip-grad-norm")
                 if sch.get(k) is not None}
 
+    def _fedavg_groups(self):
+        """Same-(cluster, stage) client groups for the RCCL all-reduce FedAvg
+        path: every rank creates these communicators in this exact order, and
+        each group's members all-reduce weight*size / size instead of shipping
+        state dicts through the control plane (replaces the reference's
+        server-side averaging loop, src/Server.py:398-408)."""
+        groups = []
+        for k in range(self.num_cluster):
+            for stage in range(1, self.n_stages + 1):
+                ids = sorted(c["client_id"] for c in self.list_clients
+                             if c["train"] and c["cluster"] == k
+                             and c["layer_id"] == stage)
+                if ids:
+                    groups.append(ids)
+        return groups
+
     def _send_start(self, rec, full_state, state_override=None):
         layers = self._stage_layers(rec)
         state = state_override if state_override is not None             else self._slice_state(full_state, layers)
@@ -305,6 +321,8 @@ class Server:
             "refresh": self.refresh, "cluster": rec["cluster"],
             "n_stages": self.n_stages, "routing": self._routing_for(rec),
             "scheduler": self.scheduler_overrides(rec),
+            "fedavg": (self.config.get("transport") or {}).get("fedavg", "control"),
+            "fedavg_groups": self._fedavg_groups(),
         })
 
     def _send_stop(self, rec, message="Stop training!"):

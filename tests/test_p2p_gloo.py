@@ -18,9 +18,10 @@ def _free_port():
     return p
 
 
-def _cfg(tmpdir, clients, cuts, num_sample=48, batch=16):
+def _cfg(tmpdir, clients, cuts, num_sample=48, batch=16, fedavg="control"):
     from split_learning_amd.config import load_config
     return load_config(None, overrides={
+        "transport": {"fedavg": fedavg},
         "server": {
             "global-round": 1, "clients": clients, "model": "VGG16",
             "data-name": "CIFAR10", "validation": False,
@@ -37,12 +38,13 @@ def _cfg(tmpdir, clients, cuts, num_sample=48, batch=16):
     })
 
 
-def _worker(rank, world, pg_port, ctl_port, tmpdir, clients, cuts):
+def _worker(rank, world, pg_port, ctl_port, tmpdir, clients, cuts,
+            fedavg="control"):
     import torch.distributed as dist
     from split_learning_amd.parallel.launch import run_p2p_client
     dist.init_process_group("gloo", init_method=f"tcp://127.0.0.1:{pg_port}",
                             rank=rank, world_size=world)
-    cfg = _cfg(tmpdir, clients, cuts)
+    cfg = _cfg(tmpdir, clients, cuts, fedavg=fedavg)
     run_p2p_client(cfg, rank, world, torch.device("cpu"), "127.0.0.1", ctl_port,
                    checkpoint_dir=str(tmpdir))
     dist.barrier()
@@ -71,3 +73,20 @@ def test_p2p_three_stage_round(tmp_path):
                             [7, 14]),
              nprocs=world, join=True)
     assert os.path.exists(os.path.join(str(tmp_path), "VGG16_CIFAR10.pth"))
+
+
+@pytest.mark.timeout(900)
+def test_p2p_rccl_allreduce_fedavg(tmp_path):
+    """2+1 clients with the all-reduce FedAvg path: the stage-1 pair averages
+    its parameters over a collective group; only the representative ships the
+    dict; the saved checkpoint has the full key set."""
+    world = 3
+    pg_port, ctl_port = _free_port(), _free_port()
+    mp.spawn(_worker, args=(world, pg_port, ctl_port, str(tmp_path), [2, 1], [7],
+                            "rccl"),
+             nprocs=world, join=True)
+    ckpt = os.path.join(str(tmp_path), "VGG16_CIFAR10.pth")
+    assert os.path.exists(ckpt)
+    sd = torch.load(ckpt, weights_only=True)
+    from split_learning_amd.models import get_model_class
+    assert set(sd.keys()) == set(get_model_class("VGG16", "CIFAR10")().state_dict().keys())
