@@ -54,26 +54,47 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
 
   f32x4 acc[2][2] = {};
 
-  for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
-    // stage A[k][m] (k-fast gather for contiguous global rows)
+  // register-prefetch double buffering (CDNA guide G15 async-STAGE split):
+  // each K-tile's gathers are ISSUED right after the previous tile's LDS
+  // write + barrier, so their global latency hides under the MFMA phase —
+  // the gather path is latency-bound at low wave counts without this.
+  constexpr int RA = (SLK_BM * SLK_BK) / 256;  // per-thread A elements
+  constexpr int RB = (SLK_BN * SLK_BK) / 256;  // per-thread B elements
+  float ra[RA], rb[RB];
+
+  auto load_tile = [&](int k0) {
     #pragma unroll
-    for (int i = 0; i < (SLK_BM * SLK_BK) / 256; ++i) {
+    for (int i = 0; i < RA; ++i) {
       int idx = tid + i * 256;
-      int m = idx >> 4;          // 0..63
-      int k = idx & 15;          // 0..15
-      ldsA[k][m] = (k0 + k < k_end && m0 + m < M) ? g.loadA(batch, m0 + m, k0 + k)
-                                                  : 0.0f;
+      int m = idx >> 4;          // 0..63 (k-fast: contiguous global rows)
+      int k = idx & 15;
+      ra[i] = (k0 + k < k_end && m0 + m < M) ? g.loadA(batch, m0 + m, k0 + k)
+                                             : 0.0f;
     }
-    // stage B[k][n] (n-fast: coalesced for row-major B)
     #pragma unroll
-    for (int i = 0; i < (SLK_BN * SLK_BK) / 256; ++i) {
+    for (int i = 0; i < RB; ++i) {
       int idx = tid + i * 256;
-      int k = idx >> 6;          // 0..15
-      int n = idx & 63;          // 0..63
-      ldsB[k][n] = (k0 + k < k_end && n0 + n < N) ? g.loadB(batch, k0 + k, n0 + n)
-                                                  : 0.0f;
+      int k = idx >> 6;          // 0..15 (n-fast: coalesced for row-major B)
+      int n = idx & 63;
+      rb[i] = (k0 + k < k_end && n0 + n < N) ? g.loadB(batch, k0 + k, n0 + n)
+                                             : 0.0f;
+    }
+  };
+
+  load_tile(k_begin);
+  for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
+    #pragma unroll
+    for (int i = 0; i < RA; ++i) {
+      int idx = tid + i * 256;
+      ldsA[idx & 15][idx >> 4] = ra[i];
+    }
+    #pragma unroll
+    for (int i = 0; i < RB; ++i) {
+      int idx = tid + i * 256;
+      ldsB[idx >> 6][idx & 63] = rb[i];
     }
     __syncthreads();
+    if (k0 + SLK_BK < k_end) load_tile(k0 + SLK_BK);  // overlaps the MFMAs
 
     #pragma unroll
     for (int kk = 0; kk < SLK_BK / 4; ++kk) {
