@@ -82,18 +82,17 @@ class ColocatedPipeline:
         return torch.nn.functional.cross_entropy(logits, labels)
 
     def _step(self):
-        with torch.no_grad():
-            act = self.s1_model(self.x_buf)
-        act_in = act.detach().requires_grad_(True)
-        self.s2_opt.zero_grad()
+        # colocated stages share the device, so there is never more than one
+        # microbatch in flight: stash the stage-1 graph instead of recomputing
+        # (identical math at control-count 1; one stage-1 forward saved)
+        out1 = self.s1_model(self.x_buf)
+        act_in = out1.detach().requires_grad_(True)
         logits = self.s2_model(act_in)
         loss = self._ce(logits, self.y_buf)
         self.nan_flag |= torch.isnan(loss)
         loss.backward()
         self.s2_opt.step()
-        self.s1_opt.zero_grad()
-        out = self.s1_model(self.x_buf)
-        out.backward(gradient=act_in.grad)
+        out1.backward(gradient=act_in.grad)
         self.s1_opt.step()
 
     def _capture(self):

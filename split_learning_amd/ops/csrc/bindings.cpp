@@ -56,6 +56,12 @@ void sgd_step(std::vector<at::Tensor>, std::vector<at::Tensor>,
 void adamw_step(std::vector<at::Tensor>, std::vector<at::Tensor>,
                 std::vector<at::Tensor>, std::vector<at::Tensor>, int64_t, double,
                 double, double, double, double, bool);
+at::Tensor make_opt_desc(std::vector<at::Tensor>, std::vector<at::Tensor>,
+                         std::vector<at::Tensor>, std::vector<at::Tensor>);
+void sgd_step_fused(const at::Tensor&, int64_t, int64_t, double, double, double,
+                    bool, bool);
+void adamw_step_fused(const at::Tensor&, int64_t, int64_t, int64_t, double, double,
+                      double, double, double, bool);
 }  // namespace slk
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -94,4 +100,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &slk::ce_bwd);
   m.def("sgd_step", &slk::sgd_step);
   m.def("adamw_step", &slk::adamw_step);
+  m.def("make_opt_desc", &slk::make_opt_desc);
+  m.def("sgd_step_fused", &slk::sgd_step_fused);
+  m.def("adamw_step_fused", &slk::adamw_step_fused);
 }

@@ -58,6 +58,131 @@ static inline int opt_grid(long n) {
   return (int)std::min<long>((n + 255) / 256, 2048);
 }
 
+// ---- fully fused multi-tensor path: ONE launch per optimizer step ----
+// desc layout (int64, device): [prefix(n+1) | numel(n) | p(n) | g(n) | b1(n) | b2(n)]
+// prefix[t] = first chunk index of tensor t; CHUNK elements per block.
+constexpr long SLK_OPT_CHUNK = 16384;
+
+__device__ __forceinline__ int find_tensor(const long* prefix, int n, long chunk) {
+  int lo = 0, hi = n;  // prefix[t] <= chunk < prefix[t+1]
+  while (hi - lo > 1) {
+    const int mid = (lo + hi) >> 1;
+    if (prefix[mid] <= chunk) lo = mid; else hi = mid;
+  }
+  return lo;
+}
+
+__global__ void sgd_fused_kernel(const long* __restrict__ desc, int n, float lr,
+                                 float momentum, float weight_decay, bool first,
+                                 bool zero_after) {
+  const long* prefix = desc;
+  const long* numels = desc + (n + 1);
+  const long* pp = numels + n;
+  const long* gp = pp + n;
+  const long* bp = gp + n;
+  const int t = find_tensor(prefix, n, blockIdx.x);
+  const long base = ((long)blockIdx.x - prefix[t]) * SLK_OPT_CHUNK;
+  const long end = min(base + SLK_OPT_CHUNK, numels[t]);
+  float* p = (float*)pp[t];
+  float* g = (float*)gp[t];
+  float* buf = (float*)bp[t];
+  for (long i = base + threadIdx.x; i < end; i += blockDim.x) {
+    float grad = g[i];
+    if (zero_after) g[i] = 0.f;
+    if (weight_decay != 0.f) grad += weight_decay * p[i];
+    float b;
+    if (momentum != 0.f) {
+      b = first ? grad : momentum * buf[i] + grad;
+      buf[i] = b;
+    } else {
+      b = grad;
+    }
+    p[i] -= lr * b;
+  }
+}
+
+__global__ void adamw_fused_kernel(const long* __restrict__ desc, int n, float lr,
+                                   float beta1, float beta2, float eps,
+                                   float weight_decay, float bc1, float bc2,
+                                   bool zero_after) {
+  const long* prefix = desc;
+  const long* numels = desc + (n + 1);
+  const long* pp = numels + n;
+  const long* gp = pp + n;
+  const long* mp = gp + n;
+  const long* vp = mp + n;
+  const int t = find_tensor(prefix, n, blockIdx.x);
+  const long base = ((long)blockIdx.x - prefix[t]) * SLK_OPT_CHUNK;
+  const long end = min(base + SLK_OPT_CHUNK, numels[t]);
+  float* p = (float*)pp[t];
+  float* g = (float*)gp[t];
+  float* m = (float*)mp[t];
+  float* v = (float*)vp[t];
+  for (long i = base + threadIdx.x; i < end; i += blockDim.x) {
+    float pv = p[i] * (1.f - lr * weight_decay);
+    const float grad = g[i];
+    if (zero_after) g[i] = 0.f;
+    const float mi = beta1 * m[i] + (1.f - beta1) * grad;
+    const float vi = beta2 * v[i] + (1.f - beta2) * grad * grad;
+    m[i] = mi;
+    v[i] = vi;
+    p[i] = pv - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+  }
+}
+
+at::Tensor make_opt_desc(std::vector<at::Tensor> params,
+                         std::vector<at::Tensor> grads,
+                         std::vector<at::Tensor> s1,
+                         std::vector<at::Tensor> s2) {
+  const int n = (int)params.size();
+  std::vector<long> host((n + 1) + n * 5);
+  long* prefix = host.data();
+  long* numels = prefix + (n + 1);
+  long* pp = numels + n;
+  long* gp = pp + n;
+  long* b1 = gp + n;
+  long* b2 = b1 + n;
+  long chunks = 0;
+  for (int i = 0; i < n; ++i) {
+    prefix[i] = chunks;
+    const long ne = params[i].numel();
+    numels[i] = ne;
+    chunks += (ne + SLK_OPT_CHUNK - 1) / SLK_OPT_CHUNK;
+    pp[i] = (long)params[i].data_ptr<float>();
+    gp[i] = (long)grads[i].data_ptr<float>();
+    b1[i] = (long)s1[i].data_ptr<float>();
+    b2[i] = s2.empty() ? 0 : (long)s2[i].data_ptr<float>();
+  }
+  prefix[n] = chunks;
+  auto t = at::from_blob(host.data(), {(long)host.size()}, at::kLong).clone();
+  return t.to(params[0].device());
+}
+
+int64_t opt_desc_chunks(const at::Tensor& desc, int64_t n) {
+  return desc[ n ].item<int64_t>();
+}
+
+void sgd_step_fused(const at::Tensor& desc, int64_t n, int64_t chunks, double lr,
+                    double momentum, double weight_decay, bool first,
+                    bool zero_after) {
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(sgd_fused_kernel, dim3((uint32_t)chunks), dim3(256), 0, stream,
+                     desc.data_ptr<int64_t>(), (int)n, (float)lr, (float)momentum,
+                     (float)weight_decay, first, zero_after);
+}
+
+void adamw_step_fused(const at::Tensor& desc, int64_t n, int64_t chunks,
+                      int64_t step, double lr, double beta1, double beta2,
+                      double eps, double weight_decay, bool zero_after) {
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const float bc1 = 1.f - powf((float)beta1, (float)step);
+  const float bc2 = 1.f - powf((float)beta2, (float)step);
+  hipLaunchKernelGGL(adamw_fused_kernel, dim3((uint32_t)chunks), dim3(256), 0,
+                     stream, desc.data_ptr<int64_t>(), (int)n, (float)lr,
+                     (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                     bc1, bc2, zero_after);
+}
+
 void sgd_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
               std::vector<at::Tensor> bufs, double lr, double momentum,
               double weight_decay, bool first, bool zero_after) {

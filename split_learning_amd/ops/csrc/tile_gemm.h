@@ -145,8 +145,9 @@ inline void slk_launch_gemm(const Gather& g, const Store& st, int M, int N, int 
 // the MFMAs — measured in profiles/).
 inline int slk_pick_split_k(int M, int N, int K, int n_batch) {
   long tiles = (long)ceil_div(M, SLK_BM) * ceil_div(N, SLK_BN) * (n_batch > 0 ? n_batch : 1);
-  if (tiles >= 768 || K <= SLK_BK * 2) return 1;
-  long want = (768 + tiles - 1) / tiles;
+  if (tiles >= 384 || K <= SLK_BK * 2) return 1;
+  long want = (384 + tiles - 1) / tiles;
+  if (want > 16) want = 16;  // >16-way fp32 atomics per output serialize
   long maxk = (K + 2 * SLK_BK - 1) / (2 * SLK_BK);  // keep >=2 BK steps per split
   long sk = want < maxk ? want : maxk;
   return (int)(sk < 1 ? 1 : sk);

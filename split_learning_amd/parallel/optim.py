@@ -23,7 +23,23 @@ class FusedSGD:
         self.weight_decay = weight_decay
         self.bufs = [torch.zeros_like(p) for p in self.params]
         self.steps = 0
-        self._torch_opt = None
+        self._desc_cache = None  # (ptr_signature, desc_tensor, n, chunks)
+
+    def _desc(self, live):
+        """Cached device descriptor for the single-launch fused step; grads
+        keep stable storage because the update kernel zeroes them in place, so
+        the table survives across steps (rebuilt if any pointer moves)."""
+        sig = tuple(p.grad.data_ptr() for p, _ in live) + \
+              tuple(p.data_ptr() for p, _ in live)
+        if self._desc_cache is None or self._desc_cache[0] != sig:
+            from ..ops import native
+            params = [p for p, _ in live]
+            desc = native().make_opt_desc(params, [p.grad for p, _ in live],
+                                          [b for _, b in live],
+                                          [b for _, b in live])
+            chunks = sum((p.numel() + 16383) // 16384 for p in params)
+            self._desc_cache = (sig, desc, len(params), chunks)
+        return self._desc_cache[1], self._desc_cache[2], self._desc_cache[3]
 
     def zero_grad(self):
         """No-op by design: step() zeroes grads in the update kernel itself
@@ -36,10 +52,10 @@ class FusedSGD:
         if not live:
             return
         if live[0][0].is_cuda:
-            hf.sgd_step([p for p, _ in live], [p.grad for p, _ in live],
-                        [b for _, b in live], self.lr, self.momentum,
-                        self.weight_decay, first_step=(self.steps == 0),
-                        zero_grad_after=True)
+            desc, n, chunks = self._desc(live)
+            from ..ops import native
+            native().sgd_step_fused(desc, n, chunks, self.lr, self.momentum,
+                                    self.weight_decay, self.steps == 0, True)
         else:
             for p, buf in live:
                 g = p.grad
@@ -67,6 +83,20 @@ class FusedAdamW:
         self.m = [torch.zeros_like(p) for p in self.params]
         self.v = [torch.zeros_like(p) for p in self.params]
         self.steps = 0
+        self._desc_cache = None
+
+    def _desc(self, live):
+        sig = tuple(p.grad.data_ptr() for p, _, _ in live) + \
+              tuple(p.data_ptr() for p, _, _ in live)
+        if self._desc_cache is None or self._desc_cache[0] != sig:
+            from ..ops import native
+            params = [p for p, _, _ in live]
+            desc = native().make_opt_desc(params, [p.grad for p, _, _ in live],
+                                          [m for _, m, _ in live],
+                                          [v for _, _, v in live])
+            chunks = sum((p.numel() + 16383) // 16384 for p in params)
+            self._desc_cache = (sig, desc, len(params), chunks)
+        return self._desc_cache[1], self._desc_cache[2], self._desc_cache[3]
 
     def zero_grad(self):
         """No-op: see FusedSGD.zero_grad."""
@@ -79,10 +109,11 @@ class FusedAdamW:
         if not live:
             return
         if live[0][0].is_cuda:
-            hf.adamw_step([p for p, _, _ in live], [p.grad for p, _, _ in live],
-                          [m for _, m, _ in live], [v for _, _, v in live],
-                          self.steps, self.lr, self.beta1, self.beta2, self.eps,
-                          self.weight_decay, zero_grad_after=True)
+            desc, n, chunks = self._desc(live)
+            from ..ops import native
+            native().adamw_step_fused(desc, n, chunks, self.steps, self.lr,
+                                      self.beta1, self.beta2, self.eps,
+                                      self.weight_decay, True)
         else:
             b1, b2 = self.beta1, self.beta2
             bc1 = 1 - b1 ** self.steps

@@ -334,3 +334,48 @@ def test_kwt_forward_matches_cpu():
         y_cpu = model(x)
         y_gpu = model.cuda()(x.cuda())
     assert_close(y_gpu.cpu(), y_cpu, atol=5e-3, rtol=5e-3, what="kwt cpu-vs-gpu")
+
+
+def test_fused_sgd_class_matches_torch():
+    """FusedSGD (single-launch descriptor path) vs torch.optim.SGD over several
+    backward-driven steps on a real module."""
+    from split_learning_amd.parallel.optim import FusedSGD
+    torch.manual_seed(0)
+    m1 = torch.nn.Sequential(torch.nn.Linear(32, 64), torch.nn.ReLU(),
+                             torch.nn.Linear(64, 8)).cuda()
+    m2 = torch.nn.Sequential(torch.nn.Linear(32, 64), torch.nn.ReLU(),
+                             torch.nn.Linear(64, 8)).cuda()
+    m2.load_state_dict(m1.state_dict())
+    o1 = FusedSGD(m1.parameters(), lr=0.01, momentum=0.5)
+    o2 = torch.optim.SGD(m2.parameters(), lr=0.01, momentum=0.5)
+    for i in range(4):
+        x = torch.randn(16, 32, device="cuda")
+        y1 = m1(x).sum()
+        y2 = m2(x).sum()
+        y1.backward()
+        y2.backward()
+        o1.step()
+        o2.step()
+        o2.zero_grad()
+    for a, b in zip(m1.parameters(), m2.parameters()):
+        assert_close(a, b, atol=1e-6, rtol=1e-6, what="FusedSGD")
+        assert torch.all(a.grad == 0)  # zero_grad_after
+
+
+def test_fused_adamw_class_matches_torch():
+    from split_learning_amd.parallel.optim import FusedAdamW
+    torch.manual_seed(0)
+    m1 = torch.nn.Linear(32, 32).cuda()
+    m2 = torch.nn.Linear(32, 32).cuda()
+    m2.load_state_dict(m1.state_dict())
+    o1 = FusedAdamW(m1.parameters(), lr=1e-3, weight_decay=0.01)
+    o2 = torch.optim.AdamW(m2.parameters(), lr=1e-3, weight_decay=0.01)
+    for i in range(4):
+        x = torch.randn(8, 32, device="cuda")
+        m1(x).sum().backward()
+        m2(x).sum().backward()
+        o1.step()
+        o2.step()
+        o2.zero_grad()
+    for a, b in zip(m1.parameters(), m2.parameters()):
+        assert_close(a, b, atol=1e-5, rtol=1e-5, what="FusedAdamW")
