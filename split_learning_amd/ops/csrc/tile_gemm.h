@@ -140,14 +140,23 @@ inline void slk_launch_gemm(const Gather& g, const Store& st, int M, int N, int 
                      stream, g, st, M, N, K, split_k, k_per_split);
 }
 
-// Heuristic: pick split_k so the grid oversubscribes the 256 CUs (~3 blocks/CU
-// target: at 1 block/CU only 4 waves are resident and the gather loads stall
-// the MFMAs — measured in profiles/).
+// Heuristic: pick split_k so the grid oversubscribes the 256 CUs (~5 blocks/CU:
+// the GPU split-K sweep in profiles/ shows throughput rising monotonically to a
+// ~1280-block target — the gather path needs many resident waves, and 48-way
+// fp32 atomic accumulation costs less than the idle CUs it fills).
 inline int slk_pick_split_k(int M, int N, int K, int n_batch) {
+  // env-tunable (host-side, read per call so in-process sweeps work):
+  // SLK_SPLIT_TARGET = block-count target, SLK_SPLIT_CAP = max split factor
+  static auto readenv = [](const char* n, long d) {
+    const char* v = std::getenv(n);
+    return v ? atol(v) : d;
+  };
+  const long target = readenv("SLK_SPLIT_TARGET", 1280);
+  const long cap = readenv("SLK_SPLIT_CAP", 64);
   long tiles = (long)ceil_div(M, SLK_BM) * ceil_div(N, SLK_BN) * (n_batch > 0 ? n_batch : 1);
-  if (tiles >= 384 || K <= SLK_BK * 2) return 1;
-  long want = (384 + tiles - 1) / tiles;
-  if (want > 16) want = 16;  // >16-way fp32 atomics per output serialize
+  if (tiles >= target || K <= SLK_BK * 2) return 1;
+  long want = (target + tiles - 1) / tiles;
+  if (want > cap) want = cap;
   long maxk = (K + 2 * SLK_BK - 1) / (2 * SLK_BK);  // keep >=2 BK steps per split
   long sk = want < maxk ? want : maxk;
   return (int)(sk < 1 ? 1 : sk);
