@@ -140,7 +140,7 @@ class DistPipeline:
         self.is_first = rank < half
         self.peer = rank + half if self.is_first else rank - half
         act_shape = (BATCH, 64, 16, 16)  # VGG16 cut=7 boundary
-        gf, gb = make_p2p_groups()
+        gf, gb = make_p2p_groups(device)
         if self.is_first:
             self.model, self.opt = build_stage([0, CUT], device)
             self.plane = P2PData(rank, device, BATCH, down_peer=self.peer,

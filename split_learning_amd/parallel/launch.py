@@ -86,13 +86,25 @@ def run_loopback(config: Dict[str, Any], device: str = "cpu",
     return server, runtimes
 
 
-def make_p2p_groups():
+def make_p2p_groups(device: Optional[torch.device] = None):
     """Create the forward/backward process groups (collective: every rank must
-    call this once, in the same order, right after init_process_group)."""
+    call this once, in the same order, right after init_process_group).
+
+    NCCL initialises communicators LAZILY on first use and init is a blocking
+    collective — if stage-1 ranks first touch only the bwd group while stage-2
+    ranks first touch the fwd group, comm init cross-deadlocks.  So both comms
+    are warmed up EAGERLY here with a tiny all-reduce every rank joins."""
     import torch.distributed as dist
     world = list(range(dist.get_world_size()))
     group_fwd = dist.new_group(world)
     group_bwd = dist.new_group(world)
+    if dist.get_backend() == "nccl":
+        dev = device if device is not None else torch.device(
+            "cuda", torch.cuda.current_device())
+        t = torch.zeros(1, device=dev)
+        dist.all_reduce(t, group=group_fwd)
+        dist.all_reduce(t, group=group_bwd)
+        torch.cuda.synchronize()
     return group_fwd, group_bwd
 
 
@@ -137,7 +149,7 @@ def run_p2p_client(config: Dict[str, Any], rank: int, world: int,
             layer_id = stage
             break
         acc += n
-    group_fwd, group_bwd = make_p2p_groups()
+    group_fwd, group_bwd = make_p2p_groups(device)
     control = StoreControl.create(store_addr, store_port, is_server=(rank == 0))
     logger = logger or Logger(f"{config['log_path']}/app.log", config["debug_mode"])
 
