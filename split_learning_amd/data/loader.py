@@ -29,5 +29,6 @@ def data_loader(data_name: str, batch_size: int,
     else:
         x, y = synthetic_tensors(data_name, distribution, seed=seed)
     ds = TensorDataset(x, y)
+    gen = torch.Generator().manual_seed(seed * 7919 + 13) if train else None
     return DataLoader(ds, batch_size=batch_size, shuffle=train, drop_last=drop_last,
-                      num_workers=0)
+                      num_workers=0, generator=gen)

@@ -61,6 +61,20 @@ class SequentialUnits(PartitionedModel):
                 setattr(self, f"layer{i}", factories[i]())
 
     def forward(self, x):
-        for i in self.active_units():
-            x = getattr(self, f"layer{i}")(x)
+        # stage-executor fusion: an adjacent BatchNorm2d -> ReLU pair inside
+        # one partition runs as a single fused kernel on GPU (the cut can
+        # still split the pair, in which case both run standalone)
+        from ..ops.modules import HipBatchNorm2d, HipReLU
+        units = self.active_units()
+        n = len(units)
+        i = 0
+        while i < n:
+            mod = getattr(self, f"layer{units[i]}")
+            if (isinstance(mod, HipBatchNorm2d) and x.is_cuda and i + 1 < n
+                    and isinstance(getattr(self, f"layer{units[i + 1]}"), HipReLU)):
+                x = mod(x, fuse_relu=True)
+                i += 2
+                continue
+            x = mod(x)
+            i += 1
         return x

@@ -42,14 +42,15 @@ class HipLinear(nn.Linear):
 
 
 class HipBatchNorm2d(nn.BatchNorm2d):
-    def forward(self, x):
+    def forward(self, x, fuse_relu: bool = False):
         if _on_gpu(x):
             if self.training and self.track_running_stats and self.num_batches_tracked is not None:
                 self.num_batches_tracked.add_(1)
             return hf.batch_norm2d(x, self.weight, self.bias, self.running_mean,
                                    self.running_var, self.training, self.momentum,
-                                   self.eps, fuse_relu=False)
-        return super().forward(x)
+                                   self.eps, fuse_relu=fuse_relu)
+        y = super().forward(x)
+        return torch.relu(y) if fuse_relu else y
 
 
 class HipReLU(nn.ReLU):

@@ -32,6 +32,30 @@ import torch
 from ..ops import functional as hf
 from .messages import ActivationMsg, GradientMsg
 
+import contextlib
+import os
+
+_TRACE = os.environ.get("SL_TRACE", "0") == "1"
+
+
+def trace_range(name):
+    """roctx marker range (torch.cuda.nvtx maps to roctx on ROCm) — makes the
+    stage phases visible in rocprofv3 timelines when SL_TRACE=1."""
+    if _TRACE and torch.cuda.is_available():
+        return _NvtxRange(name)
+    return contextlib.nullcontext()
+
+
+class _NvtxRange:
+    def __init__(self, name):
+        self.name = name
+
+    def __enter__(self):
+        torch.cuda.nvtx.range_push(self.name)
+
+    def __exit__(self, *a):
+        torch.cuda.nvtx.range_pop()
+
 
 def _cross_entropy(logits, labels):
     if logits.is_cuda:
@@ -108,13 +132,14 @@ def train_first_stage(ctx: StageContext):
         g = plane.recv_gradient(ctx.layer_id, ctx.client_id, block=False)
         if g is not None:
             stashed = inflight.pop(g.data_id)
-            opt.zero_grad()
-            if ctx.recompute:
-                out = model(stashed)          # stashed = input batch
-            else:
-                out = stashed                 # stashed = output w/ live graph
-            out.backward(gradient=g.data.to(ctx.device, non_blocking=True))
-            opt.step()
+            with trace_range("stage1.backward"):
+                opt.zero_grad()
+                if ctx.recompute:
+                    out = model(stashed)          # stashed = input batch
+                else:
+                    out = stashed                 # stashed = output w/ live graph
+                out.backward(gradient=g.data.to(ctx.device, non_blocking=True))
+                opt.step()
             n_bwd += 1
             if ctx.on_step is not None:
                 ctx.on_step(n_bwd)
@@ -128,11 +153,12 @@ def train_first_stage(ctx: StageContext):
                 end_data = True
                 continue
             x = x.to(ctx.device, non_blocking=True)
-            if ctx.recompute:
-                with torch.no_grad():
+            with trace_range("stage1.forward"):
+                if ctx.recompute:
+                    with torch.no_grad():
+                        out = model(x)
+                else:
                     out = model(x)
-            else:
-                out = model(x)
             data_id = next_id
             next_id += 1
             inflight[data_id] = x if ctx.recompute else out
