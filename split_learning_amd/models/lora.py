@@ -33,8 +33,15 @@ class LoRALinear(nn.Module):
 
     def forward(self, x):
         y = self.base(x)
-        xa = self.lora_dropout(x) @ self.lora_A.t()
-        return y + (xa @ self.lora_B.t()) * self.scaling
+        xd = self.lora_dropout(x)
+        if x.is_cuda:
+            from ..ops import functional as hf
+            xa = hf.linear(xd, self.lora_A, None)
+            delta = hf.linear(xa, self.lora_B, None)
+        else:
+            xa = xd @ self.lora_A.t()
+            delta = xa @ self.lora_B.t()
+        return y + delta * self.scaling
 
     def merged_weight(self) -> torch.Tensor:
         return self.base.weight + (self.lora_B @ self.lora_A) * self.scaling

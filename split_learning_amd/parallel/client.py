@@ -83,10 +83,16 @@ class ClientRuntime:
             "out_cluster": self.out_cluster,
         })
 
+    CONTROL_TIMEOUT_S = 600.0
+
     def run(self, max_batches: Optional[int] = None, on_step=None):
         """Main loop: handle START/SYN/STOP until the server stops us."""
         while True:
-            msg = self.inbox.recv(block=True)
+            msg = self.inbox.recv(block=True, timeout=self.CONTROL_TIMEOUT_S)
+            if msg is None:
+                raise TimeoutError(
+                    f"client {self.client_id}: no control message within "
+                    f"{self.CONTROL_TIMEOUT_S}s — server lost?")
             action = msg.get("action")
             if action == "START":
                 self._handle_start(msg)

@@ -1,0 +1,78 @@
+#!/usr/bin/env python3
+"""Secondary per-model step benchmarks (1 GPU, colocated split pipeline):
+BERT/AGNEWS cut=2 (the reference's measured BERT cut), KWT cut=7,
+MobileNetv1 cut=40, ViT cut=6 — samples/sec forward+backward+optimizer on the
+native kernel path.  Evidence that the whole model zoo runs and performs."""
+
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from split_learning_amd.models import build_partition  # noqa: E402
+from split_learning_amd.models.lora import apply_lora  # noqa: E402
+from split_learning_amd.parallel.optim import make_optimizer  # noqa: E402
+from split_learning_amd.data.synthetic import SHAPES  # noqa: E402
+from split_learning_amd.ops import functional as hf  # noqa: E402
+
+LEARNING = {"learning-rate": 5e-4, "weight-decay": 0.01, "momentum": 0.5}
+
+
+def bench_model(model, data, cut, batch, steps=64, warmup=16, lora=False):
+    dev = torch.device("cuda:0")
+    torch.manual_seed(0)
+    s1 = build_partition(model, data, [0, cut]).to(dev).train()
+    s2 = build_partition(model, data, [cut, -1]).to(dev).train()
+    if lora:
+        apply_lora(s1)
+        apply_lora(s2, trainable_extra=(f"layer{s2.TOTAL_UNITS}.classifier",))
+    o1 = make_optimizer(model, s1.parameters(), LEARNING)
+    o2 = make_optimizer(model, s2.parameters(), LEARNING)
+    shape, n_labels, dtype, vocab = SHAPES[data]
+    if dtype == torch.int64:
+        x = torch.randint(1, vocab, (batch, *shape), device=dev)
+    else:
+        x = torch.randn(batch, *shape, device=dev)
+    y = torch.randint(0, n_labels, (batch,), device=dev)
+
+    def step():
+        out1 = s1(x)
+        act = out1.detach().requires_grad_(True)
+        loss = hf.cross_entropy(s2(act), y)
+        loss.backward()
+        o2.step()
+        out1.backward(gradient=act.grad)
+        o1.step()
+        return loss
+
+    for _ in range(warmup):
+        step()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        loss = step()
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    return {
+        "model": f"{model}/{data}", "cut": cut, "batch": batch,
+        "ms_per_step": round(dt / steps * 1e3, 3),
+        "samples_per_sec": round(steps * batch / dt, 1),
+        "lora": lora, "loss": round(float(loss.detach()), 4),
+    }
+
+
+if __name__ == "__main__":
+    results = []
+    for args in [("VGG16", "CIFAR10", 7, 32, False),
+                 ("BERT", "AGNEWS", 2, 32, True),
+                 ("KWT", "SPEECHCOMMANDS", 7, 32, False),
+                 ("MobileNetv1", "CIFAR10", 40, 32, False),
+                 ("ViT", "CIFAR10", 6, 32, False)]:
+        m, d, c, b, lora = args
+        r = bench_model(m, d, c, b, lora=lora)
+        results.append(r)
+        print(json.dumps(r), flush=True)
