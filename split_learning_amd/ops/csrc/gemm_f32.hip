@@ -105,6 +105,10 @@ at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
   // y[M,N] = x[M,K] @ w[N,K]^T + b; split-K when the tile grid underfills the
   // chip (batch-32 rows = 1 M-tile: the 4096->10 classifier is ONE tile).
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat, "linear_fwd: fp32 GPU");
+  TORCH_CHECK(w.is_cuda() && w.scalar_type() == at::kFloat,
+              "linear_fwd: weight must be a fp32 GPU tensor (got device ",
+              w.device(), ")");
+  TORCH_CHECK(!bias.has_value() || bias->is_cuda(), "linear_fwd: bias device");
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1));
   const int M = x.size(0), N = w.size(0), K = x.size(1);
   int split_k = slk_pick_split_k(M, N, K, 1);

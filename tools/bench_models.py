@@ -30,6 +30,8 @@ def bench_model(model, data, cut, batch, steps=64, warmup=16, lora=False):
     if lora:
         apply_lora(s1)
         apply_lora(s2, trainable_extra=(f"layer{s2.TOTAL_UNITS}.classifier",))
+        s1.to(dev)
+        s2.to(dev)  # LoRA adds fresh CPU parameters
     o1 = make_optimizer(model, s1.parameters(), LEARNING)
     o2 = make_optimizer(model, s2.parameters(), LEARNING)
     shape, n_labels, dtype, vocab = SHAPES[data]
