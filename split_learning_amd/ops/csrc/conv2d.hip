@@ -5,16 +5,17 @@
 // Bwd-weight:  gw[Co, (Ci,KH,KW)] = gy-gather @ im2col(x)^T
 //
 // Performance notes (profiled on MI355X, profiles/):
-// * gather index math is the hot cost at these shapes, so the kernel geometry
-//   (KH, KW, stride, pad) is a TEMPLATE specialisation for the model zoo's
-//   cases — 3x3 s1 p1, 3x3 s2 p1, 1x1, 4x4 s4 (ViT patch) — making every
-//   division strength-reduced; runtime dims (OW, OH*OW, H*W) divide through
-//   FastDiv magics (common.h);
-// * every conv GEMM split-Ks when its tile grid underfills the 256 CUs
-//   (deep VGG layers have N = B*OH*OW as small as 128), with fp32 atomic
-//   accumulation and first-split-gated bias;
-// * covers every conv in the zoo (reference conv sites:
-//   src/model/VGG16_CIFAR10.py:10-94, MobileNetv1, ViT patch embed).
+// * kernel geometry (KH, KW, stride, pad) is a TEMPLATE specialisation for the
+//   model zoo's cases — 3x3 s1 p1, 3x3 s2 p1, 1x1, 4x4 s4 (ViT patch) — so
+//   every k-decomposition division is strength-reduced; runtime dims (OW,
+//   OH*OW, H*W) divide through FastDiv magics (common.h);
+// * gathers follow the tile framework's prep/load contract: the n/m-side
+//   decomposition is hoisted into per-thread contexts, and the per-element
+//   loads are BRANCHLESS (clamped addresses + cndmask selects) — the earlier
+//   bounds-branch version spent 53% of wave cycles issue-stalled behind
+//   s_and_saveexec chains;
+// * every conv GEMM split-Ks when its tile grid underfills the chip, with
+//   fp32 atomic accumulation and first-split-gated bias.
 #include <torch/extension.h>
 #include <ATen/ATen.h>
 #include <c10/hip/HIPStream.h>
@@ -23,20 +24,45 @@
 
 namespace slk {
 
+// empty + async memset instead of at::zeros: the fill kernel launches were
+// ~60/step in the profile; the memset path is cheaper and graph-capturable.
+static inline at::Tensor zeroed(at::IntArrayRef sizes, const at::TensorOptions& opt) {
+  auto t = at::empty(sizes, opt);
+  HIP_CHECK(hipMemsetAsync(t.data_ptr(), 0, t.numel() * t.element_size(),
+                           c10::hip::getCurrentHIPStream().stream()));
+  return t;
+}
+
+
 struct ConvGeom {
   int B, Ci, H, W, Co, KH, KW, OH, OW, stride, pad;
   FastDiv d_ohow, d_ow, d_hw, d_w, d_khkw, d_kw;  // runtime-dim magics
 };
 
-// compile-time geometry: CK=0 means runtime (generic fallback)
+// compile-time geometry: CKH=0 means runtime (generic fallback)
 template <int CKH, int CKW, int CS, int CP>
 struct Geo {
   static constexpr bool fixed = CKH > 0;
   __device__ static int kh_kw(const ConvGeom& g) { return fixed ? CKH * CKW : g.KH * g.KW; }
-  __device__ static int kw(const ConvGeom& g) { return fixed ? CKW : g.KW; }
   __device__ static int stride(const ConvGeom& g) { return fixed ? CS : g.stride; }
   __device__ static int pad(const ConvGeom& g) { return fixed ? CP : g.pad; }
+  // k -> (c, kh, kw) with strength-reduced division when fixed
+  __device__ static void dk(const ConvGeom& g, int k, int& c, int& kh, int& kw) {
+    if (fixed) {
+      c = k / (CKH * CKW);
+      const int r = k - c * (CKH * CKW);
+      kh = r / CKW;
+      kw = r - kh * CKW;
+    } else {
+      c = g.d_khkw.div(k);
+      const int r = g.d_khkw.mod(k, c);
+      kh = g.d_kw.div(r);
+      kw = g.d_kw.mod(r, kh);
+    }
+  }
 };
+
+__device__ __forceinline__ float sel0(float v, bool keep) { return keep ? v : 0.f; }
 
 // ---------------- forward ----------------
 template <int CKH, int CKW, int CS, int CP>
@@ -45,30 +71,35 @@ struct ConvFwdGather {
   const float* w;  // [Co, Ci, KH, KW]
   const float* x;  // [B, Ci, H, W]
   ConvGeom geo;
-  __device__ float loadA(int, int m, int k) const {  // A[Co][Ci*KH*KW]
-    return w[(long)m * (geo.Ci * G::kh_kw(geo)) + k];
+
+  struct ACtx { const float* row; bool valid; };
+  struct BCtx { const float* base; int ihb, iwb; bool valid; };
+
+  __device__ ACtx prepA(int, int m, bool valid) const {
+    return {w + (long)m * (geo.Ci * G::kh_kw(geo)), valid};
   }
-  __device__ float loadB(int, int k, int n) const {  // B[Ci*KH*KW][B*OH*OW]
+  __device__ float loadA(const ACtx& c, int k, bool kv) const {
+    return sel0(c.row[k], c.valid & kv);
+  }
+  __device__ BCtx prepB(int, int n, bool valid) const {
     const unsigned b = geo.d_ohow.div(n);
     const unsigned rem = geo.d_ohow.mod(n, b);
     const unsigned oh = geo.d_ow.div(rem);
     const unsigned ow = geo.d_ow.mod(rem, oh);
+    return {x + (long)b * geo.Ci * geo.H * geo.W,
+            (int)oh * G::stride(geo) - G::pad(geo),
+            (int)ow * G::stride(geo) - G::pad(geo), valid};
+  }
+  __device__ float loadB(const BCtx& c, int k, bool kv) const {
     int ci, kh, kw;
-    if (G::fixed) {
-      ci = k / (CKH * CKW);
-      const int r = k - ci * (CKH * CKW);
-      kh = r / CKW;
-      kw = r - kh * CKW;
-    } else {
-      ci = geo.d_khkw.div(k);
-      const int r = geo.d_khkw.mod(k, ci);
-      kh = geo.d_kw.div(r);
-      kw = geo.d_kw.mod(r, kh);
-    }
-    const int ih = (int)oh * G::stride(geo) - G::pad(geo) + kh;
-    const int iw = (int)ow * G::stride(geo) - G::pad(geo) + kw;
-    if (ih < 0 || ih >= geo.H || iw < 0 || iw >= geo.W) return 0.f;
-    return x[(((long)b * geo.Ci + ci) * geo.H + ih) * geo.W + iw];
+    G::dk(geo, k, ci, kh, kw);
+    const int ih = c.ihb + kh;
+    const int iw = c.iwb + kw;
+    const bool in = (unsigned)ih < (unsigned)geo.H && (unsigned)iw < (unsigned)geo.W;
+    const int ihc = in ? ih : 0;
+    const int iwc = in ? iw : 0;
+    const float v = c.base[((long)ci * geo.H + ihc) * geo.W + iwc];
+    return sel0(v, c.valid & kv & in);
   }
 };
 
@@ -98,49 +129,56 @@ struct ConvBwdDataGather {
   const float* w;   // [Co, Ci, KH, KW]
   const float* gy;  // [B, Co, OH, OW]
   ConvGeom geo;
-  __device__ float loadA(int, int m, int k) const {  // A[Ci][Co*KH*KW]
+
+  struct ACtx { int m; bool valid; };
+  struct BCtx { const float* base; int ihp, iwp; bool valid; };
+
+  __device__ ACtx prepA(int, int m, bool valid) const { return {m, valid}; }
+  __device__ float loadA(const ACtx& c, int k, bool kv) const {
+    int co, kh, kw;
+    G::dk(geo, k, co, kh, kw);
     const int khkw = G::kh_kw(geo);
-    int co, r;
-    if (G::fixed) {
-      co = k / (CKH * CKW);
-      r = k - co * (CKH * CKW);
-    } else {
-      co = geo.d_khkw.div(k);
-      r = geo.d_khkw.mod(k, co);
-    }
-    return w[((long)co * geo.Ci + m) * khkw + r];
+    const float v = w[((long)co * geo.Ci + c.m) * khkw + kh * (G::fixed ? CKW : geo.KW) + kw];
+    return sel0(v, c.valid & kv);
   }
-  __device__ float loadB(int, int k, int n) const {  // B[Co*KH*KW][B*H*W]
+  __device__ BCtx prepB(int, int n, bool valid) const {
     const unsigned b = geo.d_hw.div(n);
     const unsigned rem = geo.d_hw.mod(n, b);
     const unsigned ih = geo.d_w.div(rem);
     const unsigned iw = geo.d_w.mod(rem, ih);
+    return {gy + (long)b * geo.Co * geo.OH * geo.OW,
+            (int)ih + G::pad(geo), (int)iw + G::pad(geo), valid};
+  }
+  __device__ float loadB(const BCtx& c, int k, bool kv) const {
     int co, kh, kw;
-    if (G::fixed) {
-      co = k / (CKH * CKW);
-      const int r = k - co * (CKH * CKW);
-      kh = r / CKW;
-      kw = r - kh * CKW;
-    } else {
-      co = geo.d_khkw.div(k);
-      const int r = geo.d_khkw.mod(k, co);
-      kh = geo.d_kw.div(r);
-      kw = geo.d_kw.mod(r, kh);
-    }
+    G::dk(geo, k, co, kh, kw);
     const int s = G::stride(geo);
-    const int oh_num = (int)ih + G::pad(geo) - kh;
-    const int ow_num = (int)iw + G::pad(geo) - kw;
+    const int oh_num = c.ihp - kh;
+    const int ow_num = c.iwp - kw;
     int oh, ow;
+    bool ok;
     if (s == 1) {
       oh = oh_num;
       ow = ow_num;
+      ok = true;
+    } else if (s == 2) {
+      ok = ((oh_num | ow_num) & 1) == 0;
+      oh = oh_num >> 1;
+      ow = ow_num >> 1;
+    } else if (s == 4) {
+      ok = ((oh_num | ow_num) & 3) == 0;
+      oh = oh_num >> 2;
+      ow = ow_num >> 2;
     } else {
-      if ((oh_num % s) != 0 || (ow_num % s) != 0) return 0.f;
+      ok = (oh_num % s) == 0 && (ow_num % s) == 0;
       oh = oh_num / s;
       ow = ow_num / s;
     }
-    if (oh < 0 || ow < 0 || oh >= geo.OH || ow >= geo.OW) return 0.f;
-    return gy[(((long)b * geo.Co + co) * geo.OH + oh) * geo.OW + ow];
+    ok = ok && (unsigned)oh < (unsigned)geo.OH && (unsigned)ow < (unsigned)geo.OW;
+    const int ohc = ok ? oh : 0;
+    const int owc = ok ? ow : 0;
+    const float v = c.base[((long)co * geo.OH + ohc) * geo.OW + owc];
+    return sel0(v, c.valid & kv & ok);
   }
 };
 
@@ -168,32 +206,36 @@ struct ConvBwdWeightGather {
   const float* gy;  // [B, Co, OH, OW]
   const float* x;   // [B, Ci, H, W]
   ConvGeom geo;
-  __device__ float loadA(int, int m, int k) const {  // A[Co][B*OH*OW]
+
+  struct ACtx { int m; bool valid; };
+  struct BCtx { long ciHW; int kh, kw; bool valid; };
+
+  __device__ ACtx prepA(int, int m, bool valid) const { return {m, valid}; }
+  __device__ float loadA(const ACtx& c, int k, bool kv) const {
+    // k = (b, oh, ow): runtime-size decomposition via FastDiv
     const unsigned b = geo.d_ohow.div(k);
     const unsigned rem = geo.d_ohow.mod(k, b);
-    return gy[((long)b * geo.Co + m) * (geo.OH * geo.OW) + rem];
+    const float v = gy[((long)b * geo.Co + c.m) * (geo.OH * geo.OW) + rem];
+    return sel0(v, c.valid & kv);
   }
-  __device__ float loadB(int, int k, int n) const {  // B[B*OH*OW][Ci*KH*KW]
+  __device__ BCtx prepB(int, int n, bool valid) const {
+    int ci, kh, kw;
+    G::dk(geo, n, ci, kh, kw);
+    return {(long)ci * geo.H * geo.W, kh, kw, valid};
+  }
+  __device__ float loadB(const BCtx& c, int k, bool kv) const {
     const unsigned b = geo.d_ohow.div(k);
     const unsigned rem = geo.d_ohow.mod(k, b);
     const unsigned oh = geo.d_ow.div(rem);
     const unsigned ow = geo.d_ow.mod(rem, oh);
-    int ci, kh, kw;
-    if (G::fixed) {
-      ci = n / (CKH * CKW);
-      const int r = n - ci * (CKH * CKW);
-      kh = r / CKW;
-      kw = r - kh * CKW;
-    } else {
-      ci = geo.d_khkw.div(n);
-      const int r = geo.d_khkw.mod(n, ci);
-      kh = geo.d_kw.div(r);
-      kw = geo.d_kw.mod(r, kh);
-    }
-    const int ih = (int)oh * G::stride(geo) - G::pad(geo) + kh;
-    const int iw = (int)ow * G::stride(geo) - G::pad(geo) + kw;
-    if (ih < 0 || ih >= geo.H || iw < 0 || iw >= geo.W) return 0.f;
-    return x[(((long)b * geo.Ci + ci) * geo.H + ih) * geo.W + iw];
+    const int ih = (int)oh * G::stride(geo) - G::pad(geo) + c.kh;
+    const int iw = (int)ow * G::stride(geo) - G::pad(geo) + c.kw;
+    const bool in = (unsigned)ih < (unsigned)geo.H && (unsigned)iw < (unsigned)geo.W;
+    const int ihc = in ? ih : 0;
+    const int iwc = in ? iw : 0;
+    const float v = x[(long)b * geo.Ci * geo.H * geo.W + c.ciHW
+                      + (long)ihc * geo.W + iwc];
+    return sel0(v, c.valid & kv & in);
   }
 };
 
@@ -263,7 +305,7 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   const int M = geo.Co, N = geo.B * geo.OH * geo.OW, K = geo.Ci * geo.KH * geo.KW;
   const int split_k = slk_pick_split_k(M, N, K, 1);
   auto y = split_k > 1
-      ? at::zeros({geo.B, geo.Co, geo.OH, geo.OW}, x.options())
+      ? zeroed({geo.B, geo.Co, geo.OH, geo.OW}, x.options())
       : at::empty({geo.B, geo.Co, geo.OH, geo.OW}, x.options());
 
   ConvFwdStore st{y.data_ptr<float>(),
@@ -287,7 +329,7 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
                            w.size(3), stride, pad);
   const int M = geo.Ci, N = geo.B * geo.H * geo.W, K = geo.Co * geo.KH * geo.KW;
   const int split_k = slk_pick_split_k(M, N, K, 1);
-  auto gx = split_k > 1 ? at::zeros({geo.B, geo.Ci, geo.H, geo.W}, gy.options())
+  auto gx = split_k > 1 ? zeroed({geo.B, geo.Ci, geo.H, geo.W}, gy.options())
                         : at::empty({geo.B, geo.Ci, geo.H, geo.W}, gy.options());
 
   ConvBwdDataStore st{gx.data_ptr<float>(), geo.Ci, geo.H * geo.W, geo.d_hw,
@@ -309,7 +351,7 @@ at::Tensor conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x, int KH,
   ConvGeom geo = make_geom(x.size(0), x.size(1), x.size(2), x.size(3), gy.size(1),
                            KH, KW, stride, pad);
   const int M = geo.Co, N = geo.Ci * KH * KW, K = geo.B * geo.OH * geo.OW;
-  auto gw = at::zeros({geo.Co, geo.Ci, KH, KW}, gy.options());
+  auto gw = zeroed({geo.Co, geo.Ci, KH, KW}, gy.options());
 
   AtomicStore st{gw.data_ptr<float>(), N};
   int split_k = slk_pick_split_k(M, N, K, 1);

@@ -8,10 +8,19 @@
 //
 // Geometry: 256 threads = 4 waves; block tile 64(M) x 64(N); BK=16 K-step;
 // each wave owns a 32x32 sub-tile = 2x2 MFMA fragments with 4 f32x4
-// accumulators.  A and B tiles are staged k-major in LDS ([BK][64+pad]) so the
-// MFMA operand reads (lane&15 consecutive) are conflict-light.  Split-K over
-// grid.z with atomic C accumulation is supported for K-heavy shapes
-// (conv bwd-weight: K = B*OH*OW).
+// accumulators.  Staging is register-prefetch double-buffered (next K-tile's
+// gathers issue under the MFMA phase).
+//
+// Gather contract (PMC-driven redesign — the first version's per-element
+// bounds branches compiled to 56 s_and_saveexec chains per loop and put waves
+// 53% issue-stalled, profiles/):
+//   * prepA(batch, m_clamped, valid) / prepB(batch, n_clamped, valid) run ONCE
+//     per thread (m/n are staging-loop invariants): they hoist the row/column
+//     address decomposition into a small ctx;
+//   * loadA(ctx, k)/loadB(ctx, k) must be BRANCHLESS: always load from a
+//     clamped in-bounds address and select 0 via the valid flags (cndmask,
+//     not exec-mask branches).  k arrives already clamped to [0, K-1];
+//     k_valid covers the K tail.
 #pragma once
 
 #include "common.h"
@@ -21,12 +30,6 @@ constexpr int SLK_BN = 64;
 constexpr int SLK_BK = 16;
 constexpr int SLK_LDS_PAD = 4;
 
-// Each gather functor provides:
-//   __device__ float loadA(int batch, int m, int k) const;  // 0 if OOB
-//   __device__ float loadB(int batch, int k, int n) const;
-// The store functor provides:
-//   __device__ void store(int batch, int m, int n, float v, bool first_split)
-//     const;  // handles OOB; first_split gates one-time epilogue work (bias)
 template <typename Gather, typename Store>
 __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     Gather g, Store st, int M, int N, int K, int split_k, int k_per_split) {
@@ -54,30 +57,39 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
 
   f32x4 acc[2][2] = {};
 
-  // register-prefetch double buffering (CDNA guide G15 async-STAGE split):
-  // each K-tile's gathers are ISSUED right after the previous tile's LDS
-  // write + barrier, so their global latency hides under the MFMA phase —
-  // the gather path is latency-bound at low wave counts without this.
   constexpr int RA = (SLK_BM * SLK_BK) / 256;  // per-thread A elements
   constexpr int RB = (SLK_BN * SLK_BK) / 256;  // per-thread B elements
   float ra[RA], rb[RB];
 
+  // hoisted per-thread staging contexts (m/n fixed across the K loop)
+  typename Gather::ACtx actx[RA];
+  typename Gather::BCtx bctx[RB];
+  int ka[RA], kb[RB];
+  #pragma unroll
+  for (int i = 0; i < RA; ++i) {
+    const int idx = tid + i * 256;
+    const int m = m0 + (idx >> 4);      // k-fast: contiguous global rows
+    ka[i] = idx & 15;
+    actx[i] = g.prepA(batch, min(m, M - 1), m < M);
+  }
+  #pragma unroll
+  for (int i = 0; i < RB; ++i) {
+    const int idx = tid + i * 256;
+    const int n = n0 + (idx & 63);      // n-fast: coalesced for row-major B
+    kb[i] = idx >> 6;
+    bctx[i] = g.prepB(batch, min(n, N - 1), n < N);
+  }
+
   auto load_tile = [&](int k0) {
     #pragma unroll
     for (int i = 0; i < RA; ++i) {
-      int idx = tid + i * 256;
-      int m = idx >> 4;          // 0..63 (k-fast: contiguous global rows)
-      int k = idx & 15;
-      ra[i] = (k0 + k < k_end && m0 + m < M) ? g.loadA(batch, m0 + m, k0 + k)
-                                             : 0.0f;
+      const int k = k0 + ka[i];
+      ra[i] = g.loadA(actx[i], min(k, K - 1), k < k_end);
     }
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
-      int idx = tid + i * 256;
-      int k = idx >> 6;          // 0..15 (n-fast: coalesced for row-major B)
-      int n = idx & 63;
-      rb[i] = (k0 + k < k_end && n0 + n < N) ? g.loadB(batch, k0 + k, n0 + n)
-                                             : 0.0f;
+      const int k = k0 + kb[i];
+      rb[i] = g.loadB(bctx[i], min(k, K - 1), k < k_end);
     }
   };
 
@@ -85,12 +97,12 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
   for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
     #pragma unroll
     for (int i = 0; i < RA; ++i) {
-      int idx = tid + i * 256;
+      const int idx = tid + i * 256;
       ldsA[idx & 15][idx >> 4] = ra[i];
     }
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
-      int idx = tid + i * 256;
+      const int idx = tid + i * 256;
       ldsB[idx >> 6][idx & 63] = rb[i];
     }
     __syncthreads();
