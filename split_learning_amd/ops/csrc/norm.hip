@@ -14,6 +14,8 @@ namespace slk {
 // empty + async memset instead of at::zeros: the fill kernel launches were
 // ~60/step in the profile; the memset path is cheaper and graph-capturable.
 static inline at::Tensor zeroed(at::IntArrayRef sizes, const at::TensorOptions& opt) {
+  static const bool use_memset = std::getenv("SLK_NO_MEMSET") == nullptr;
+  if (!use_memset) return at::zeros(sizes, opt);
   auto t = at::empty(sizes, opt);
   HIP_CHECK(hipMemsetAsync(t.data_ptr(), 0, t.numel() * t.element_size(),
                            c10::hip::getCurrentHIPStream().stream()));
