@@ -49,10 +49,24 @@ def build_stage(layers, device):
     return model, opt
 
 
+_LABEL_PROJ = None
+
+
 def make_batches(device, n, seed=0):
+    """Synthetic CIFAR10-shaped batches.  Labels are a fixed random projection
+    of the image (argmax of P @ x): a learnable target with the same shapes
+    and compute cost as real labels.  Pure-random labels put BN-training in a
+    pathological regime (dead-channel variance collapse amplifies gradients
+    ~invstd ≈ 300x and intermittently diverges to NaN after a few hundred
+    steps — measured; real CIFAR10 does not behave that way)."""
+    global _LABEL_PROJ
     g = torch.Generator(device="cpu").manual_seed(seed)
     x = torch.randn(n, BATCH, 3, 32, 32, generator=g).to(device)
-    y = torch.randint(0, 10, (n, BATCH), generator=g).to(device)
+    if _LABEL_PROJ is None:
+        gp = torch.Generator(device="cpu").manual_seed(777)
+        _LABEL_PROJ = torch.randn(3 * 32 * 32, 10, generator=gp).to(device)
+    with torch.no_grad():
+        y = (x.reshape(n * BATCH, -1) @ _LABEL_PROJ).argmax(-1).reshape(n, BATCH)
     return x, y
 
 
