@@ -52,13 +52,91 @@ def _load_cifar10(train: bool):
     return torch.from_numpy(x), torch.tensor(ys, dtype=torch.int64)
 
 
+def _load_mnist(train: bool):
+    """Raw idx files (train-images-idx3-ubyte etc.) under DATA_ROOT/MNIST/raw."""
+    base = os.path.join(DATA_ROOT, "MNIST", "raw")
+    stem = "train" if train else "t10k"
+    img_p = os.path.join(base, f"{stem}-images-idx3-ubyte")
+    lbl_p = os.path.join(base, f"{stem}-labels-idx1-ubyte")
+    if not (os.path.exists(img_p) and os.path.exists(lbl_p)):
+        return None
+    with open(img_p, "rb") as f:
+        data = np.frombuffer(f.read(), dtype=np.uint8, offset=16)
+    with open(lbl_p, "rb") as f:
+        labels = np.frombuffer(f.read(), dtype=np.uint8, offset=8)
+    x = data.reshape(-1, 1, 28, 28).astype(np.float32) / 255.0
+    x = (x - 0.1307) / 0.3081
+    return torch.from_numpy(x.copy()), torch.tensor(labels, dtype=torch.int64)
+
+
+def _load_agnews(train: bool):
+    """AG_NEWS csv (class,title,description) tokenised with BertTokenizer to
+    max_len 128 — reference src/dataset/AGNEWS.py / dataloader.py:16-59."""
+    p = os.path.join(DATA_ROOT, "ag_news", "train.csv" if train else "test.csv")
+    if not os.path.exists(p):
+        return None
+    import csv as _csv
+    from transformers import BertTokenizer
+    tok = BertTokenizer.from_pretrained("bert-base-cased")
+    ids, labels = [], []
+    with open(p, newline="") as f:
+        for row in _csv.reader(f):
+            label = int(row[0]) - 1
+            text = " ".join(row[1:])
+            enc = tok(text, max_length=128, truncation=True, padding="max_length")
+            ids.append(enc["input_ids"])
+            labels.append(label)
+    return (torch.tensor(ids, dtype=torch.int64),
+            torch.tensor(labels, dtype=torch.int64))
+
+
+_SC_CLASSES = ["yes", "no", "up", "down", "left", "right", "on", "off", "stop", "go"]
+
+
+def _load_speechcommands(train: bool, max_per_class: int = 400):
+    """SpeechCommands v0.02 wavs -> numpy MFCC [40, 98]
+    (reference src/dataset/SPEECHCOMMANDS.py; frontend in data/mfcc.py)."""
+    base = os.path.join(DATA_ROOT, "SpeechCommands", "speech_commands_v0.02")
+    if not os.path.isdir(base):
+        return None
+    from scipy.io import wavfile
+    from .mfcc import compute_mfcc
+    test_set = set()
+    tl = os.path.join(base, "testing_list.txt")
+    if os.path.exists(tl):
+        test_set = set(open(tl).read().split())
+    xs, ys = [], []
+    for ci, cls in enumerate(_SC_CLASSES):
+        d = os.path.join(base, cls)
+        if not os.path.isdir(d):
+            continue
+        count = 0
+        for fn in sorted(os.listdir(d)):
+            if not fn.endswith(".wav") or count >= max_per_class:
+                continue
+            rel = f"{cls}/{fn}"
+            if train == (rel in test_set):
+                continue
+            _sr, wav = wavfile.read(os.path.join(d, fn))
+            wav = wav.astype(np.float32) / 32768.0
+            if len(wav) < 16000:
+                wav = np.pad(wav, (0, 16000 - len(wav)))
+            xs.append(compute_mfcc(wav[:16000]).astype(np.float32))
+            ys.append(ci)
+            count += 1
+    if not xs:
+        return None
+    return torch.from_numpy(np.stack(xs)), torch.tensor(ys, dtype=torch.int64)
+
+
 def load_real(data_name: str, distribution: Optional[List[int]],
               train: bool) -> Optional[Tuple[torch.Tensor, torch.Tensor]]:
-    if data_name == "CIFAR10":
-        loaded = _load_cifar10(train)
-        if loaded is None:
-            return None
-        return _subset_by_distribution(*loaded, distribution)
-    # MNIST / AGNEWS / SPEECHCOMMANDS fall back to synthetic unless local files
-    # are wired in a future round.
-    return None
+    loaders = {"CIFAR10": _load_cifar10, "MNIST": _load_mnist,
+               "AGNEWS": _load_agnews, "SPEECHCOMMANDS": _load_speechcommands}
+    fn = loaders.get(data_name)
+    if fn is None:
+        return None
+    loaded = fn(train)
+    if loaded is None:
+        return None
+    return _subset_by_distribution(*loaded, distribution)
