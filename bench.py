@@ -140,6 +140,73 @@ class ColocatedPipeline:
         return self.nan_flag
 
 
+class OverlapPipeline:
+    """N=1 path with genuine 1F1B overlap: the two colocated stages run on
+    separate HIP streams — stage-1 forward of microbatch i+1 (and the deferred
+    stage-1 backward of i-1) overlap stage-2's fwd+bwd+step of microbatch i.
+    Weight staleness is bounded by the in-flight depth, exactly the
+    reference's control-count pipelining semantics (src/train/VGG16.py:75-119).
+    Autograd replays each op on the stream its forward ran on, so stage-1
+    backward stays on sA and stage-2 on sB; cross-stream hand-offs (cut
+    activation down, cut gradient up) are event-ordered."""
+
+    def __init__(self, device, depth=CONTROL_COUNT):
+        from split_learning_amd.ops import functional as hf
+        self.hf = hf
+        self.device = device
+        self.depth = depth
+        self.s1_model, self.s1_opt = build_stage([0, CUT], device)
+        self.s2_model, self.s2_opt = build_stage([CUT, -1], device)
+        self.sA = torch.cuda.Stream()
+        self.sB = torch.cuda.Stream()
+        self.nan_flag = torch.zeros((), dtype=torch.bool, device=device)
+        self.batch_seed = 0
+
+    def _drain_one(self, pending):
+        out1, act, ev_b = pending.popleft()
+        with torch.cuda.stream(self.sA):
+            self.sA.wait_event(ev_b)
+            out1.backward(gradient=act.grad)
+            self.s1_opt.step()
+
+    def run(self, steps):
+        import collections
+        self.batch_seed += 1
+        xs, ys = make_batches(self.device, steps, seed=self.batch_seed)
+        cur = torch.cuda.current_stream()
+        self.sA.wait_stream(cur)
+        self.sB.wait_stream(cur)
+        pending = collections.deque()
+        for i in range(steps):
+            with torch.cuda.stream(self.sA):
+                out1 = self.s1_model(xs[i])
+                ev_f = torch.cuda.Event()
+                ev_f.record(self.sA)
+            with torch.cuda.stream(self.sB):
+                self.sB.wait_event(ev_f)
+                act = out1.detach().requires_grad_(True)
+                logits = self.s2_model(act)
+                loss = self._ce(logits, ys[i])
+                self.nan_flag |= torch.isnan(loss)
+                loss.backward()
+                self.s2_opt.step()
+                ev_b = torch.cuda.Event()
+                ev_b.record(self.sB)
+            pending.append((out1, act, ev_b))
+            if len(pending) >= self.depth:
+                self._drain_one(pending)
+        while pending:
+            self._drain_one(pending)
+        cur.wait_stream(self.sA)
+        cur.wait_stream(self.sB)
+        return self.nan_flag
+
+    def _ce(self, logits, labels):
+        if logits.is_cuda:
+            return self.hf.cross_entropy(logits, labels)
+        return torch.nn.functional.cross_entropy(logits, labels)
+
+
 class DistPipeline:
     """One rank per GPU; pair (r, r + world/2) forms a pipeline.  Uses the
     production P2PData plane (pre-posted irecv rings, fwd/bwd on separate
@@ -230,6 +297,9 @@ def main():
                          "investigation.")
     ap.add_argument("--no-graphs", action="store_true",
                     help="(compat) force-disable graph capture")
+    ap.add_argument("--serial", action="store_true",
+                    help="N=1: serial colocated step instead of the 1F1B "
+                         "two-stream overlap")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -265,6 +335,9 @@ def main():
 
     if dist_mode:
         pipeline = DistPipeline(rank, world, device)
+    elif have_gpu and not args.serial:
+        # default N=1 mode: 1F1B stage overlap on two HIP streams
+        pipeline = OverlapPipeline(device)
     else:
         pipeline = ColocatedPipeline(device, use_graphs=args.graphs and not args.no_graphs)
     runner = pipeline.run
