@@ -166,6 +166,7 @@ class OverlapPipeline:
         out1, act, ev_b = pending.popleft()
         with torch.cuda.stream(self.sA):
             self.sA.wait_event(ev_b)
+            act.grad.record_stream(self.sA)  # grad was allocated on sB
             out1.backward(gradient=act.grad)
             self.s1_opt.step()
 
@@ -184,6 +185,7 @@ class OverlapPipeline:
                 ev_f.record(self.sA)
             with torch.cuda.stream(self.sB):
                 self.sB.wait_event(ev_f)
+                out1.record_stream(self.sB)  # activation storage came from sA
                 act = out1.detach().requires_grad_(True)
                 logits = self.s2_model(act)
                 loss = self._ce(logits, ys[i])

@@ -86,7 +86,18 @@ class ClientRuntime:
     CONTROL_TIMEOUT_S = 600.0
 
     def run(self, max_batches: Optional[int] = None, on_step=None):
-        """Main loop: handle START/SYN/STOP until the server stops us."""
+        """Main loop: handle START/SYN/STOP until the server stops us.
+
+        On a GPU, each client thread gets its OWN HIP stream so colocated
+        stages (loopback mode: several clients sharing one device) genuinely
+        overlap — the loopback data plane orders cross-stream tensor hand-offs
+        with events."""
+        if self.device.type == "cuda":
+            with torch.cuda.stream(torch.cuda.Stream(device=self.device)):
+                return self._run_loop(max_batches, on_step)
+        return self._run_loop(max_batches, on_step)
+
+    def _run_loop(self, max_batches=None, on_step=None):
         while True:
             msg = self.inbox.recv(block=True, timeout=self.CONTROL_TIMEOUT_S)
             if msg is None:

@@ -42,7 +42,13 @@ MAX_TRACE = 8  # max pipeline depth for the trace stack
 
 
 class LoopbackData:
-    """Single-process data plane; queue per activation edge / gradient target."""
+    """Single-process data plane; queue per activation edge / gradient target.
+
+    Tensors pass by reference (zero-copy).  When clients run on per-thread HIP
+    streams of ONE GPU (colocated stages overlap like the N=1 bench), each
+    message carries a CUDA event recorded on the sender's stream; the receiver
+    waits on it before touching the tensor — cross-stream hand-off is
+    event-ordered, never implicit."""
 
     def __init__(self):
         self._act: Dict[Tuple[int, int], queue.Queue] = {}
@@ -55,28 +61,54 @@ class LoopbackData:
                 table[key] = queue.Queue()
             return table[key]
 
+    @staticmethod
+    def _wrap(msg):
+        if msg.data is not None and msg.data.is_cuda:
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+            return (msg, ev)
+        return (msg, None)
+
+    @staticmethod
+    def _unwrap(item):
+        msg, ev = item
+        if ev is not None:
+            cur = torch.cuda.current_stream()
+            cur.wait_event(ev)
+            # the consumer stream now uses storage the sender's stream
+            # allocated: tell the caching allocator so the block is not
+            # recycled by the sender before these kernels finish
+            if msg.data is not None and msg.data.is_cuda:
+                msg.data.record_stream(cur)
+            labels = getattr(msg, "labels", None)
+            if labels is not None and labels.is_cuda:
+                labels.record_stream(cur)
+        return msg
+
     def send_activation(self, stage: int, cluster: int, msg: ActivationMsg,
                         dst_client: Optional[int] = None):
-        self._q(self._act, (stage, cluster)).put(msg)
+        self._q(self._act, (stage, cluster)).put(self._wrap(msg))
 
     def recv_activation(self, prev_stage: int, cluster: int, my_client: int,
                         block=False, timeout=None) -> Optional[ActivationMsg]:
         try:
-            return self._q(self._act, (prev_stage, cluster)).get(block=block,
+            item = self._q(self._act, (prev_stage, cluster)).get(block=block,
                                                                  timeout=timeout)
         except queue.Empty:
             return None
+        return self._unwrap(item)
 
     def send_gradient(self, stage: int, to_client: int, msg: GradientMsg):
-        self._q(self._grad, (stage, to_client)).put(msg)
+        self._q(self._grad, (stage, to_client)).put(self._wrap(msg))
 
     def recv_gradient(self, stage: int, client: int, block=False,
                       timeout=None) -> Optional[GradientMsg]:
         try:
-            return self._q(self._grad, (stage, client)).get(block=block,
+            item = self._q(self._grad, (stage, client)).get(block=block,
                                                             timeout=timeout)
         except queue.Empty:
             return None
+        return self._unwrap(item)
 
     def flush(self):
         pass
