@@ -366,28 +366,36 @@ at::Tensor conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x, int KH,
   return gw;
 }
 
-// per-channel sum of gy over (B, OH, OW) -> conv bias gradient
+// per-channel sum of gy over (B, OH, OW) -> conv bias gradient (chunked)
 __global__ void conv_bias_grad_kernel(const float* __restrict__ gy,
                                       float* __restrict__ gb, int B, int C, int HW) {
   __shared__ float scratch[16];
   const int c = blockIdx.x;
+  const int total = B * HW;
+  const int per = (total + gridDim.y - 1) / gridDim.y;
+  const int lo = blockIdx.y * per;
+  const int hi = min(total, lo + per);
   float acc = 0.f;
-  for (int i = threadIdx.x; i < B * HW; i += blockDim.x) {
+  for (int i = lo + threadIdx.x; i < hi; i += blockDim.x) {
     const int b = i / HW;
     const int r = i - b * HW;
     acc += gy[((long)b * C + c) * HW + r];
   }
-  float total = slk_block_sum(acc, scratch);
-  if (threadIdx.x == 0) gb[c] = total;
+  float total_s = slk_block_sum(acc, scratch);
+  if (threadIdx.x == 0) atomicAdd(gb + c, total_s);
 }
 
 at::Tensor conv2d_bwd_bias(const at::Tensor& gy) {
   auto gyc = gy.contiguous();
   const int B = gy.size(0), C = gy.size(1), HW = gy.size(2) * gy.size(3);
-  auto gb = at::empty({C}, gy.options());
+  auto gb = zeroed({C}, gy.options());
+  long chunks = ((long)B * HW) / 4096;
+  if (chunks < 1) chunks = 1;
+  if (chunks > 16) chunks = 16;
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(conv_bias_grad_kernel, dim3(C), dim3(256), 0, stream,
-                     gyc.data_ptr<float>(), gb.data_ptr<float>(), B, C, HW);
+  hipLaunchKernelGGL(conv_bias_grad_kernel, dim3(C, (uint32_t)chunks), dim3(256),
+                     0, stream, gyc.data_ptr<float>(), gb.data_ptr<float>(), B, C,
+                     HW);
   return gb;
 }
 

@@ -90,7 +90,9 @@ __global__ void bn_fwd_kernel(const float* __restrict__ x, float* __restrict__ y
   }
 }
 
-// reductions for backward: sum(gy) and sum(gy * xhat) per channel
+// reductions for backward: sum(gy) and sum(gy * xhat) per channel, chunked
+// over grid.y like the forward stats (one block per channel leaves most CUs
+// idle at C=64)
 __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
                                      const float* __restrict__ gy,
                                      const float* __restrict__ mean,
@@ -101,8 +103,12 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
   __shared__ double scratch[16];
   const int c = blockIdx.x;
   const float m = mean[c], is = invstd[c];
+  const int total = B * HW;
+  const int per = (total + gridDim.y - 1) / gridDim.y;
+  const int lo = blockIdx.y * per;
+  const int hi = min(total, lo + per);
   double s = 0.0, sx = 0.0;
-  for (int i = threadIdx.x; i < B * HW; i += blockDim.x) {
+  for (int i = lo + threadIdx.x; i < hi; i += blockDim.x) {
     const int b = i / HW;
     const int r = i - b * HW;
     const long off = ((long)b * C + c) * HW + r;
@@ -114,8 +120,8 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
   __syncthreads();
   double tsx = slk_block_sum(sx, scratch);
   if (threadIdx.x == 0) {
-    sum_gy[c] = (float)ts;
-    sum_gy_xhat[c] = (float)tsx;
+    atomicAdd(sum_gy + c, (float)ts);
+    atomicAdd(sum_gy_xhat + c, (float)tsx);
   }
 }
 
@@ -195,11 +201,12 @@ static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tens
                                              const at::Tensor& invstd, bool training) {
   const int B = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   const long total = x.numel();
-  auto sum_gy = at::empty({C}, x.options());
-  auto sum_gy_xhat = at::empty({C}, x.options());
+  auto sum_gy = zeroed({C}, x.options());
+  auto sum_gy_xhat = zeroed({C}, x.options());
   auto gx = at::empty_like(x);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C), dim3(256), 0, stream,
+  hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C, bn_chunks((long)B * HW)),
+                     dim3(256), 0, stream,
                      x.data_ptr<float>(), gy.data_ptr<float>(),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      sum_gy.data_ptr<float>(), sum_gy_xhat.data_ptr<float>(), B, C,

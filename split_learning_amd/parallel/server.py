@@ -402,7 +402,9 @@ class Server:
                 ok = True
                 if self.validation:
                     from ..validation import get_val
-                    ok = get_val(self.model_name, self.data_name, full, self.logger)
+                    dev = "cuda:0" if torch.cuda.is_available() else "cpu"
+                    ok = get_val(self.model_name, self.data_name, full, self.logger,
+                                 device=dev)
                 if ok:
                     torch.save(full, self.ckpt_path)
                     self.round -= 1
