@@ -22,13 +22,25 @@ import torch.nn.functional as F
 from . import functional as hf
 
 
+import os
+
+# debug bisection toggles: SLK_DBG_TORCH=conv,bn,linear,pool,dropout,relu
+# routes the named op families to stock torch GPU kernels (used to isolate the
+# graph-replay divergence; no effect unless the env var is set)
+_DBG_TORCH = frozenset((os.environ.get("SLK_DBG_TORCH") or "").split(","))
+
+
+def _use_native(kind: str, x: torch.Tensor) -> bool:
+    return x.is_cuda and kind not in _DBG_TORCH
+
+
 def _on_gpu(x: torch.Tensor) -> bool:
     return x.is_cuda
 
 
 class HipConv2d(nn.Conv2d):
     def forward(self, x):
-        if _on_gpu(x):
+        if _use_native("conv", x):
             return hf.conv2d(x, self.weight, self.bias,
                              stride=self.stride[0], padding=self.padding[0])
         return super().forward(x)
@@ -36,14 +48,14 @@ class HipConv2d(nn.Conv2d):
 
 class HipLinear(nn.Linear):
     def forward(self, x):
-        if _on_gpu(x):
+        if _use_native("linear", x):
             return hf.linear(x, self.weight, self.bias)
         return super().forward(x)
 
 
 class HipBatchNorm2d(nn.BatchNorm2d):
     def forward(self, x, fuse_relu: bool = False):
-        if _on_gpu(x):
+        if _use_native("bn", x):
             if self.training and self.track_running_stats and self.num_batches_tracked is not None:
                 self.num_batches_tracked.add_(1)
             return hf.batch_norm2d(x, self.weight, self.bias, self.running_mean,
@@ -55,7 +67,7 @@ class HipBatchNorm2d(nn.BatchNorm2d):
 
 class HipReLU(nn.ReLU):
     def forward(self, x):
-        if _on_gpu(x):
+        if _use_native("relu", x):
             return hf.relu(x)
         return super().forward(x)
 
@@ -76,14 +88,14 @@ class HipTanh(nn.Tanh):
 
 class HipMaxPool2d(nn.MaxPool2d):
     def forward(self, x):
-        if _on_gpu(x) and self.kernel_size == 2 and self.stride == 2:
+        if _use_native("pool", x) and self.kernel_size == 2 and self.stride == 2:
             return hf.maxpool2x2(x)
         return super().forward(x)
 
 
 class HipDropout(nn.Dropout):
     def forward(self, x):
-        if _on_gpu(x):
+        if _use_native("dropout", x):
             return hf.dropout(x, self.p, self.training)
         return super().forward(x)
 
