@@ -21,6 +21,22 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 
 static inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
 
+// Zero-fill as a KERNEL node, never hipMemsetAsync: captured memset nodes were
+// observed to intermittently misorder against dependent kernel nodes on
+// hipGraph replay (per-process at graphExec instantiation; root-caused by
+// bisection — see profiles/SUMMARY.md "graph divergence").  Kernel nodes
+// order correctly; cost is the same single launch.
+static __global__ void slk_zero_kernel(float* __restrict__ p, long n) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+    p[i] = 0.f;
+}
+
+static inline void slk_zero_async(float* p, long n, hipStream_t stream) {
+  const int grid = (int)std::min<long>((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(slk_zero_kernel, dim3(grid), dim3(256), 0, stream, p, n);
+}
+
 // splitmix64 — counter-based RNG hash for dropout (deterministic per
 // (seed, offset, index); quality is ample for Bernoulli masks).
 __device__ __forceinline__ uint64_t slk_mix64(uint64_t x) {

@@ -10,14 +10,12 @@
 
 namespace slk {
 
-// empty + async memset instead of at::zeros: the fill kernel launches were
-// ~60/step in the profile; the memset path is cheaper and graph-capturable.
+// empty + single fill-KERNEL launch instead of at::zeros (no aten dispatch;
+// and never hipMemsetAsync — see slk_zero_async in common.h).
 static inline at::Tensor zeroed(at::IntArrayRef sizes, const at::TensorOptions& opt) {
-  static const bool use_memset = std::getenv("SLK_NO_MEMSET") == nullptr;
-  if (!use_memset) return at::zeros(sizes, opt);
   auto t = at::empty(sizes, opt);
-  HIP_CHECK(hipMemsetAsync(t.data_ptr(), 0, t.numel() * t.element_size(),
-                           c10::hip::getCurrentHIPStream().stream()));
+  slk_zero_async(t.data_ptr<float>(), t.numel(),
+                 c10::hip::getCurrentHIPStream().stream());
   return t;
 }
 
@@ -119,8 +117,8 @@ at::Tensor matmul_f32(const at::Tensor& a, const at::Tensor& b, bool ta, bool tb
   if (!accumulate && !out.has_value()) {
     split_k = slk_pick_split_k(M, N, Ka, a.dim() == 3 ? a.size(0) : 1);
     if (split_k > 1) {
-      HIP_CHECK(hipMemsetAsync(c.data_ptr(), 0, c.numel() * c.element_size(),
-                               c10::hip::getCurrentHIPStream().stream()));
+      slk_zero_async(c.data_ptr<float>(), c.numel(),
+                     c10::hip::getCurrentHIPStream().stream());
     }
   }
   launch_strided(a, b, c, ta, tb, c10::nullopt, accumulate, split_k);

@@ -11,14 +11,12 @@
 
 namespace slk {
 
-// empty + async memset instead of at::zeros: the fill kernel launches were
-// ~60/step in the profile; the memset path is cheaper and graph-capturable.
+// empty + single fill-KERNEL launch instead of at::zeros (no aten dispatch;
+// and never hipMemsetAsync — see slk_zero_async in common.h).
 static inline at::Tensor zeroed(at::IntArrayRef sizes, const at::TensorOptions& opt) {
-  static const bool use_memset = std::getenv("SLK_NO_MEMSET") == nullptr;
-  if (!use_memset) return at::zeros(sizes, opt);
   auto t = at::empty(sizes, opt);
-  HIP_CHECK(hipMemsetAsync(t.data_ptr(), 0, t.numel() * t.element_size(),
-                           c10::hip::getCurrentHIPStream().stream()));
+  slk_zero_async(t.data_ptr<float>(), t.numel(),
+                 c10::hip::getCurrentHIPStream().stream());
   return t;
 }
 
