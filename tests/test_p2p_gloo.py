@@ -90,3 +90,28 @@ def test_p2p_rccl_allreduce_fedavg(tmp_path):
     sd = torch.load(ckpt, weights_only=True)
     from split_learning_amd.models import get_model_class
     assert set(sd.keys()) == set(get_model_class("VGG16", "CIFAR10")().state_dict().keys())
+
+
+def _noniid_worker(rank, w, pgp, ctlp, tmpdir):
+    import torch.distributed as dist
+    from split_learning_amd.parallel.launch import run_p2p_client
+    dist.init_process_group("gloo", init_method=f"tcp://127.0.0.1:{pgp}",
+                            rank=rank, world_size=w)
+    cfg = _cfg(tmpdir, [2, 2], [7], num_sample=32, batch=8, fedavg="rccl")
+    cfg["server"]["data-distribution"]["non-iid"] = True
+    cfg["server"]["data-distribution"]["dirichlet"]["alpha"] = 0.5
+    run_p2p_client(cfg, rank, w, torch.device("cpu"), "127.0.0.1", ctlp,
+                   checkpoint_dir=tmpdir)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(900)
+def test_p2p_noniid_allreduce_2plus2(tmp_path):
+    """BASELINE config-3 shape scaled to CI: 2+2 clients, Dirichlet non-IID,
+    all-reduce FedAvg, 4 gloo ranks."""
+    world = 4
+    pg_port, ctl_port = _free_port(), _free_port()
+    mp.spawn(_noniid_worker, args=(world, pg_port, ctl_port, str(tmp_path)),
+             nprocs=world, join=True)
+    assert os.path.exists(os.path.join(str(tmp_path), "VGG16_CIFAR10.pth"))
