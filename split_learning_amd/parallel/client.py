@@ -203,6 +203,12 @@ class ClientRuntime:
             sync_first=bool(sch.get("sync-first", False)),
             sda_size=int(sch.get("sda-size", 1)),
             on_step=on_step,
+            # reference prints the loss each batch (src/train/VGG16.py:168);
+            # we keep that behind debug_mode since .item() syncs the device
+            log_loss=((lambda loss: self.logger.log_debug(
+                f"client {self.client_id} loss: {float(loss.detach()):.6f}"))
+                if (self.logger is not None and getattr(self.logger, "debug_mode", False)
+                    and self.layer_id == self.n_stages) else None),
         )
         result, size = run_stage(ctx)
         pause_msg = ctx.pause_msg
