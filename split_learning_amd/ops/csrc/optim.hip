@@ -82,22 +82,39 @@ __global__ void sgd_fused_kernel(const long* __restrict__ desc, int n, float lr,
   const long* bp = gp + n;
   const int t = find_tensor(prefix, n, blockIdx.x);
   const long base = ((long)blockIdx.x - prefix[t]) * SLK_OPT_CHUNK;
-  const long end = min(base + SLK_OPT_CHUNK, numels[t]);
+  const long ne = numels[t];
+  const long end = min(base + SLK_OPT_CHUNK, ne);
   float* p = (float*)pp[t];
   float* g = (float*)gp[t];
   float* buf = (float*)bp[t];
-  for (long i = base + threadIdx.x; i < end; i += blockDim.x) {
-    float grad = g[i];
+  auto upd = [&](float pv, float gv, float bv, float& po, float& bo) {
+    if (weight_decay != 0.f) gv += weight_decay * pv;
+    float b = (momentum != 0.f) ? (first ? gv : momentum * bv + gv) : gv;
+    bo = b;
+    po = pv - lr * b;
+  };
+  // float4 main body (chunk bases are 16384-element aligned; torch GPU
+  // allocations are >=256B aligned) + scalar tail
+  const long end4 = base + ((end - base) & ~3L);
+  for (long i = base + (long)threadIdx.x * 4; i < end4; i += (long)blockDim.x * 4) {
+    float4 pv = *(float4*)(p + i);
+    float4 gv = *(float4*)(g + i);
+    float4 bv = momentum != 0.f ? *(float4*)(buf + i) : float4{0, 0, 0, 0};
+    float4 po, bo;
+    upd(pv.x, gv.x, bv.x, po.x, bo.x);
+    upd(pv.y, gv.y, bv.y, po.y, bo.y);
+    upd(pv.z, gv.z, bv.z, po.z, bo.z);
+    upd(pv.w, gv.w, bv.w, po.w, bo.w);
+    *(float4*)(p + i) = po;
+    if (momentum != 0.f) *(float4*)(buf + i) = bo;
+    if (zero_after) *(float4*)(g + i) = float4{0, 0, 0, 0};
+  }
+  for (long i = end4 + threadIdx.x; i < end; i += blockDim.x) {
+    float po, bo;
+    upd(p[i], g[i], momentum != 0.f ? buf[i] : 0.f, po, bo);
+    p[i] = po;
+    if (momentum != 0.f) buf[i] = bo;
     if (zero_after) g[i] = 0.f;
-    if (weight_decay != 0.f) grad += weight_decay * p[i];
-    float b;
-    if (momentum != 0.f) {
-      b = first ? grad : momentum * buf[i] + grad;
-      buf[i] = b;
-    } else {
-      b = grad;
-    }
-    p[i] -= lr * b;
   }
 }
 
@@ -105,6 +122,8 @@ __global__ void adamw_fused_kernel(const long* __restrict__ desc, int n, float l
                                    float beta1, float beta2, float eps,
                                    float weight_decay, float bc1, float bc2,
                                    bool zero_after) {
+  // (scalar loop: AdamW tensors in this zoo are small; the SGD path above is
+  // the hot one and is vectorized)
   const long* prefix = desc;
   const long* numels = desc + (n + 1);
   const long* pp = numels + n;

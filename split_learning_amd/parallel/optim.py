@@ -55,7 +55,12 @@ class FusedSGD:
             desc, n, chunks = self._desc(live)
             from ..ops import native
             native().sgd_step_fused(desc, n, chunks, self.lr, self.momentum,
-                                    self.weight_decay, self.steps == 0, True)
+                                    self.weight_decay, self.steps == 0, False)
+            # release grads: the next backward then MOVES fresh gradients in
+            # (no accumulate-add kernel per tensor); the allocator usually
+            # hands back the same blocks, so the descriptor cache still hits
+            for p, _ in live:
+                p.grad = None
         else:
             for p, buf in live:
                 g = p.grad
@@ -113,7 +118,9 @@ class FusedAdamW:
             from ..ops import native
             native().adamw_step_fused(desc, n, chunks, self.steps, self.lr,
                                       self.beta1, self.beta2, self.eps,
-                                      self.weight_decay, True)
+                                      self.weight_decay, False)
+            for p, _, _ in live:
+                p.grad = None
         else:
             b1, b2 = self.beta1, self.beta2
             bc1 = 1 - b1 ** self.steps
