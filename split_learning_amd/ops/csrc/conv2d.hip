@@ -364,9 +364,11 @@ at::Tensor conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x, int KH,
   return gw;
 }
 
-// per-channel sum of gy over (B, OH, OW) -> conv bias gradient (chunked)
+// per-channel sum of gy over (B, OH, OW) -> conv bias gradient
+// (chunk slabs + finalize reduce: no zero-init, no atomics)
 __global__ void conv_bias_grad_kernel(const float* __restrict__ gy,
-                                      float* __restrict__ gb, int B, int C, int HW) {
+                                      float* __restrict__ slab, int B, int C,
+                                      int HW) {
   __shared__ float scratch[16];
   const int c = blockIdx.x;
   const int total = B * HW;
@@ -380,20 +382,34 @@ __global__ void conv_bias_grad_kernel(const float* __restrict__ gy,
     acc += gy[((long)b * C + c) * HW + r];
   }
   float total_s = slk_block_sum(acc, scratch);
-  if (threadIdx.x == 0) atomicAdd(gb + c, total_s);
+  if (threadIdx.x == 0) slab[(long)blockIdx.y * C + c] = total_s;
+}
+
+__global__ void conv_bias_finalize_kernel(const float* __restrict__ slab,
+                                          int chunks, float* __restrict__ gb,
+                                          int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f;
+  for (int k = 0; k < chunks; ++k) s += slab[(long)k * C + c];
+  gb[c] = s;
 }
 
 at::Tensor conv2d_bwd_bias(const at::Tensor& gy) {
   auto gyc = gy.contiguous();
   const int B = gy.size(0), C = gy.size(1), HW = gy.size(2) * gy.size(3);
-  auto gb = zeroed({C}, gy.options());
+  auto gb = at::empty({C}, gy.options());
   long chunks = ((long)B * HW) / 4096;
   if (chunks < 1) chunks = 1;
   if (chunks > 16) chunks = 16;
+  auto slab = at::empty({chunks, C}, gy.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   hipLaunchKernelGGL(conv_bias_grad_kernel, dim3(C, (uint32_t)chunks), dim3(256),
-                     0, stream, gyc.data_ptr<float>(), gb.data_ptr<float>(), B, C,
+                     0, stream, gyc.data_ptr<float>(), slab.data_ptr<float>(), B, C,
                      HW);
+  hipLaunchKernelGGL(conv_bias_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256),
+                     0, stream, slab.data_ptr<float>(), (int)chunks,
+                     gb.data_ptr<float>(), C);
   return gb;
 }
 

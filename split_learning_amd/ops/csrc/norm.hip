@@ -27,8 +27,10 @@ static inline at::Tensor zeroed(at::IntArrayRef sizes, const at::TensorOptions& 
 // by a finalize kernel that also folds invstd and the running-stat update
 // in-kernel (the previous host-side rsqrt/mul_/add_ chain was 5 extra
 // kernel launches per BN call).
+// slab layout [2][chunks][C]: every slot plainly written (no zero-init, no
+// atomics); the finalize kernel reduces over chunks
 __global__ void bn_partial_kernel(const float* __restrict__ x,
-                                  float* __restrict__ sum, float* __restrict__ sumsq,
+                                  float* __restrict__ slab,
                                   int B, int C, int HW) {
   __shared__ double scratch[16];
   const int c = blockIdx.x;
@@ -48,13 +50,13 @@ __global__ void bn_partial_kernel(const float* __restrict__ x,
   __syncthreads();
   double ts2 = slk_block_sum(s2, scratch);
   if (threadIdx.x == 0) {
-    atomicAdd(sum + c, (float)ts);
-    atomicAdd(sumsq + c, (float)ts2);
+    const long chunks = gridDim.y;
+    slab[(long)blockIdx.y * C + c] = (float)ts;
+    slab[chunks * C + (long)blockIdx.y * C + c] = (float)ts2;
   }
 }
 
-__global__ void bn_finalize_kernel(const float* __restrict__ sum,
-                                   const float* __restrict__ sumsq,
+__global__ void bn_finalize_kernel(const float* __restrict__ slab, int chunks,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
@@ -62,8 +64,13 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
                                    float momentum, float eps) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  const float m = sum[c] / n;
-  const float v = fmaxf(sumsq[c] / n - m * m, 0.f);
+  float s = 0.f, s2 = 0.f;
+  for (int k = 0; k < chunks; ++k) {
+    s += slab[(long)k * C + c];
+    s2 += slab[(long)chunks * C + (long)k * C + c];
+  }
+  const float m = s / n;
+  const float v = fmaxf(s2 / n - m * m, 0.f);
   mean[c] = m;
   invstd[c] = rsqrtf(v + eps);
   if (running_mean != nullptr) {
@@ -95,8 +102,7 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
                                      const float* __restrict__ gy,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
-                                     float* __restrict__ sum_gy,
-                                     float* __restrict__ sum_gy_xhat, int B, int C,
+                                     float* __restrict__ slab, int B, int C,
                                      int HW) {
   __shared__ double scratch[16];
   const int c = blockIdx.x;
@@ -118,9 +124,24 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
   __syncthreads();
   double tsx = slk_block_sum(sx, scratch);
   if (threadIdx.x == 0) {
-    atomicAdd(sum_gy + c, (float)ts);
-    atomicAdd(sum_gy_xhat + c, (float)tsx);
+    const long chunks = gridDim.y;
+    slab[(long)blockIdx.y * C + c] = (float)ts;
+    slab[chunks * C + (long)blockIdx.y * C + c] = (float)tsx;
   }
+}
+
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ slab, int chunks,
+                                       float* __restrict__ sum_gy,
+                                       float* __restrict__ sum_gy_xhat, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, sx = 0.f;
+  for (int k = 0; k < chunks; ++k) {
+    s += slab[(long)k * C + c];
+    sx += slab[(long)chunks * C + (long)k * C + c];
+  }
+  sum_gy[c] = s;
+  sum_gy_xhat[c] = sx;
 }
 
 __global__ void bn_bwd_dx_kernel(const float* __restrict__ x,
@@ -160,14 +181,13 @@ std::vector<at::Tensor> bn2d_stats_fused(const at::Tensor& x,
   const int B = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto mean = at::empty({C}, x.options());
   auto invstd = at::empty({C}, x.options());
-  auto sums = zeroed({2, C}, x.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int chunks = bn_chunks((long)B * HW);
+  auto slab = at::empty({2, chunks, C}, x.options());
   hipLaunchKernelGGL(bn_partial_kernel, dim3(C, chunks), dim3(256), 0, stream,
-                     x.data_ptr<float>(), sums.data_ptr<float>(),
-                     sums.data_ptr<float>() + C, B, C, HW);
+                     x.data_ptr<float>(), slab.data_ptr<float>(), B, C, HW);
   hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
-                     stream, sums.data_ptr<float>(), sums.data_ptr<float>() + C,
+                     stream, slab.data_ptr<float>(), chunks,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      running_mean.has_value() ? running_mean->data_ptr<float>()
                                               : nullptr,
@@ -199,16 +219,20 @@ static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tens
                                              const at::Tensor& invstd, bool training) {
   const int B = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   const long total = x.numel();
-  auto sum_gy = zeroed({C}, x.options());
-  auto sum_gy_xhat = zeroed({C}, x.options());
+  auto sum_gy = at::empty({C}, x.options());
+  auto sum_gy_xhat = at::empty({C}, x.options());
   auto gx = at::empty_like(x);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C, bn_chunks((long)B * HW)),
+  const int rchunks = bn_chunks((long)B * HW);
+  auto slab = at::empty({2, rchunks, C}, x.options());
+  hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C, rchunks),
                      dim3(256), 0, stream,
                      x.data_ptr<float>(), gy.data_ptr<float>(),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     sum_gy.data_ptr<float>(), sum_gy_xhat.data_ptr<float>(), B, C,
-                     HW);
+                     slab.data_ptr<float>(), B, C, HW);
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
+                     stream, slab.data_ptr<float>(), rchunks,
+                     sum_gy.data_ptr<float>(), sum_gy_xhat.data_ptr<float>(), C);
   int grid = (int)std::min<long>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(grid), dim3(256), 0, stream,
                      x.data_ptr<float>(), gy.data_ptr<float>(),
