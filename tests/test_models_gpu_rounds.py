@@ -51,3 +51,17 @@ def test_gpu_round(tmp_path, model, data, cut):
     assert set(sd.keys()) == full_keys
     for v in sd.values():
         assert not torch.isnan(v.float()).any()
+
+
+def test_gpu_three_stage_mnist(tmp_path):
+    """BASELINE config-5 analog on one GPU: VGG16/MNIST 3-stage split
+    cuts [7, 14], middle-relay on the loopback plane with per-thread streams."""
+    cfg = _cfg(tmp_path, "VGG16", "MNIST", 7)
+    cfg["server"]["clients"] = [1, 1, 1]
+    cfg["server"]["manual"]["no-cluster"]["cut-layers"] = [7, 14]
+    server, _ = run_loopback(cfg, device="cuda:0", checkpoint_dir=str(tmp_path))
+    assert server.round == 0
+    sd = torch.load(os.path.join(str(tmp_path), "VGG16_MNIST.pth"),
+                    weights_only=True)
+    assert set(sd.keys()) == set(
+        get_model_class("VGG16", "MNIST")().state_dict().keys())
