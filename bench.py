@@ -110,6 +110,9 @@ class ColocatedPipeline:
         self.s1_opt.step()
 
     def _capture(self):
+        # stable grad/descriptor pointers are required under capture
+        self.s1_opt.release_grads = False
+        self.s2_opt.release_grads = False
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):

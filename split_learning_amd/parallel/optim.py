@@ -24,6 +24,11 @@ class FusedSGD:
         self.bufs = [torch.zeros_like(p) for p in self.params]
         self.steps = 0
         self._desc_cache = None  # (ptr_signature, desc_tensor, n, chunks)
+        # release_grads=True (default): drop p.grad after the fused step so the
+        # next backward MOVES fresh gradients (no accumulate-add kernels).
+        # Set False under hipGraph capture (stable pointers required); the
+        # update kernel then zeroes grads in place instead.
+        self.release_grads = True
 
     def _desc(self, live):
         """Cached device descriptor for the single-launch fused step; grads
@@ -55,12 +60,14 @@ class FusedSGD:
             desc, n, chunks = self._desc(live)
             from ..ops import native
             native().sgd_step_fused(desc, n, chunks, self.lr, self.momentum,
-                                    self.weight_decay, self.steps == 0, False)
-            # release grads: the next backward then MOVES fresh gradients in
-            # (no accumulate-add kernel per tensor); the allocator usually
-            # hands back the same blocks, so the descriptor cache still hits
-            for p, _ in live:
-                p.grad = None
+                                    self.weight_decay, self.steps == 0,
+                                    not self.release_grads)
+            if self.release_grads:
+                # next backward MOVES fresh gradients in (no accumulate-add
+                # kernel per tensor); the allocator usually hands back the same
+                # blocks, so the descriptor cache still hits
+                for p, _ in live:
+                    p.grad = None
         else:
             for p, buf in live:
                 g = p.grad
@@ -89,6 +96,7 @@ class FusedAdamW:
         self.v = [torch.zeros_like(p) for p in self.params]
         self.steps = 0
         self._desc_cache = None
+        self.release_grads = True  # see FusedSGD
 
     def _desc(self, live):
         sig = tuple(p.grad.data_ptr() for p, _, _ in live) + \
@@ -118,9 +126,10 @@ class FusedAdamW:
             from ..ops import native
             native().adamw_step_fused(desc, n, chunks, self.steps, self.lr,
                                       self.beta1, self.beta2, self.eps,
-                                      self.weight_decay, False)
-            for p, _, _ in live:
-                p.grad = None
+                                      self.weight_decay, not self.release_grads)
+            if self.release_grads:
+                for p, _, _ in live:
+                    p.grad = None
         else:
             b1, b2 = self.beta1, self.beta2
             bc1 = 1 - b1 ** self.steps

@@ -22,7 +22,8 @@ from split_learning_amd.ops import functional as hf  # noqa: E402
 LEARNING = {"learning-rate": 5e-4, "weight-decay": 0.01, "momentum": 0.5}
 
 
-def bench_model(model, data, cut, batch, steps=64, warmup=16, lora=False):
+def bench_model(model, data, cut, batch, steps=64, warmup=16, lora=False,
+                graphs=False):
     dev = torch.device("cuda:0")
     torch.manual_seed(0)
     s1 = build_partition(model, data, [0, cut]).to(dev).train()
@@ -41,6 +42,8 @@ def bench_model(model, data, cut, batch, steps=64, warmup=16, lora=False):
         x = torch.randn(batch, *shape, device=dev)
     y = torch.randint(0, n_labels, (batch,), device=dev)
 
+    last = {}
+
     def step():
         out1 = s1(x)
         act = out1.detach().requires_grad_(True)
@@ -49,32 +52,50 @@ def bench_model(model, data, cut, batch, steps=64, warmup=16, lora=False):
         o2.step()
         out1.backward(gradient=act.grad)
         o1.step()
+        last["loss"] = loss
         return loss
 
+    graph = None
+    if graphs:
+        # graph capture freezes grad/descriptor pointers: keep grads alive
+        # (the optimizers normally release them after the fused step)
+        o1.release_grads = False
+        o2.release_grads = False
+        sstream = torch.cuda.Stream()
+        sstream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(sstream):
+            for _ in range(3):
+                step()
+        torch.cuda.current_stream().wait_stream(sstream)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            step()
+
+    runner = (lambda: graph.replay()) if graph is not None else step
     for _ in range(warmup):
-        step()
+        runner()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(steps):
-        loss = step()
+        runner()
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     return {
         "model": f"{model}/{data}", "cut": cut, "batch": batch,
         "ms_per_step": round(dt / steps * 1e3, 3),
         "samples_per_sec": round(steps * batch / dt, 1),
-        "lora": lora, "loss": round(float(loss.detach()), 4),
+        "lora": lora, "graphs": graphs,
+        "loss": round(float(last["loss"].detach()), 4),
     }
 
 
 if __name__ == "__main__":
-    results = []
+    graphs = "--graphs" in sys.argv
     for args in [("VGG16", "CIFAR10", 7, 32, False),
                  ("BERT", "AGNEWS", 2, 32, True),
                  ("KWT", "SPEECHCOMMANDS", 7, 32, False),
                  ("MobileNetv1", "CIFAR10", 40, 32, False),
                  ("ViT", "CIFAR10", 6, 32, False)]:
         m, d, c, b, lora = args
-        r = bench_model(m, d, c, b, lora=lora)
-        results.append(r)
+        r = bench_model(m, d, c, b, lora=lora, graphs=graphs)
         print(json.dumps(r), flush=True)
