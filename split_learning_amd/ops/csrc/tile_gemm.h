@@ -88,7 +88,12 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     }
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
-      const int k = k0 + kb[i];
+      // kb[i] = (tid + i*256) >> 6 is WAVE-UNIFORM by construction (the >>6
+      // collapses the 64 lanes of a wave to its wave id); readfirstlane makes
+      // that provable so the functor's whole k-decomposition (divisions,
+      // FastDiv multiplies) compiles to SALU ops instead of per-lane VALU —
+      // the gather address math was 139 VALU per 16 MFMA in the loop body.
+      const int k = __builtin_amdgcn_readfirstlane(k0 + kb[i]);
       rb[i] = g.loadB(bctx[i], min(k, K - 1), k < k_end);
     }
   };
