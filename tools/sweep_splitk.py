@@ -29,7 +29,7 @@ def run(target, cap, steps=128, warmup=24):
 
 if __name__ == "__main__":
     # interleaved A/B rounds (guide §5.4 rule 24)
-    configs = [(256, 8), (384, 16), (768, 16), (768, 48), (1024, 48), (1536, 48)]
+    configs = [(768, 48), (1024, 64), (1280, 64), (1536, 64), (2048, 128)]
     results = {c: [] for c in configs}
     for rnd in range(2):
         for c in configs:
