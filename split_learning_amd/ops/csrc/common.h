@@ -37,14 +37,6 @@ static inline void slk_zero_async(float* p, long n, hipStream_t stream) {
   hipLaunchKernelGGL(slk_zero_kernel, dim3(grid), dim3(256), 0, stream, p, n);
 }
 
-// Last-block-done reductions (bn stats, bn bwd sums, conv bias grad) need a
-// zeroed per-channel int counter array.  The fused kernels RESET their
-// counters to 0 before exiting, so one buffer is reusable launch-to-launch;
-// thread_local because concurrent client threads run on distinct HIP streams
-// (stream order alone serialises reuse within a thread).  Declared here;
-// defined where ATen is available (norm.hip) — common.h stays torch-free.
-int* slk_counter_buf(int n, void* stream);
-
 // splitmix64 — counter-based RNG hash for dropout (deterministic per
 // (seed, offset, index); quality is ample for Bernoulli masks).
 __device__ __forceinline__ uint64_t slk_mix64(uint64_t x) {

@@ -365,13 +365,9 @@ at::Tensor conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x, int KH,
 }
 
 // per-channel sum of gy over (B, OH, OW) -> conv bias gradient
-// (chunk slabs, no zero-init/atomics on the data path; the finalize reduce is
-// fused in via the last-block-done pattern — see bn_partial_kernel in
-// norm.hip for the discipline; counters come zeroed and leave zeroed)
+// (chunk slabs + finalize reduce: no zero-init, no atomics)
 __global__ void conv_bias_grad_kernel(const float* __restrict__ gy,
-                                      float* __restrict__ slab,
-                                      int* __restrict__ counters,
-                                      float* __restrict__ gb, int B, int C,
+                                      float* __restrict__ slab, int B, int C,
                                       int HW) {
   __shared__ float scratch[16];
   const int c = blockIdx.x;
@@ -386,16 +382,17 @@ __global__ void conv_bias_grad_kernel(const float* __restrict__ gy,
     acc += gy[((long)b * C + c) * HW + r];
   }
   float total_s = slk_block_sum(acc, scratch);
-  if (threadIdx.x != 0) return;
-  const int chunks = gridDim.y;
-  slab[(long)blockIdx.y * C + c] = total_s;
-  __threadfence();
-  if (atomicAdd(&counters[c], 1) != chunks - 1) return;
-  __threadfence();
+  if (threadIdx.x == 0) slab[(long)blockIdx.y * C + c] = total_s;
+}
+
+__global__ void conv_bias_finalize_kernel(const float* __restrict__ slab,
+                                          int chunks, float* __restrict__ gb,
+                                          int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
   float s = 0.f;
   for (int k = 0; k < chunks; ++k) s += slab[(long)k * C + c];
   gb[c] = s;
-  counters[c] = 0;
 }
 
 at::Tensor conv2d_bwd_bias(const at::Tensor& gy) {
@@ -408,8 +405,11 @@ at::Tensor conv2d_bwd_bias(const at::Tensor& gy) {
   auto slab = at::empty({chunks, C}, gy.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   hipLaunchKernelGGL(conv_bias_grad_kernel, dim3(C, (uint32_t)chunks), dim3(256),
-                     0, stream, gyc.data_ptr<float>(), slab.data_ptr<float>(),
-                     slk_counter_buf(C, stream), gb.data_ptr<float>(), B, C, HW);
+                     0, stream, gyc.data_ptr<float>(), slab.data_ptr<float>(), B, C,
+                     HW);
+  hipLaunchKernelGGL(conv_bias_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256),
+                     0, stream, slab.data_ptr<float>(), (int)chunks,
+                     gb.data_ptr<float>(), C);
   return gb;
 }
 

@@ -9,21 +9,6 @@
 
 #include "common.h"
 
-// shared counter buffer for the last-block-done reductions (declared in
-// common.h; conv2d.hip's bias-grad uses it too — same .so).  Every fused
-// kernel resets its counters to 0 before exiting, so reuse needs no re-zero.
-int* slk_counter_buf(int n, void* stream) {
-  thread_local static at::Tensor buf;
-  if (!buf.defined() || buf.numel() < n) {
-    buf = at::empty({std::max(n, 1024)},
-                    at::TensorOptions().dtype(at::kInt).device(at::kCUDA));
-    // int zeros == float zeros bitwise; reuse the float fill kernel
-    slk_zero_async(reinterpret_cast<float*>(buf.data_ptr<int>()), buf.numel(),
-                   (hipStream_t)stream);
-  }
-  return buf.data_ptr<int>();
-}
-
 namespace slk {
 
 // empty + single fill-KERNEL launch instead of at::zeros (no aten dispatch;
@@ -38,23 +23,15 @@ static inline at::Tensor zeroed(at::IntArrayRef sizes, const at::TensorOptions& 
 
 // ---------------- BatchNorm2d ----------------
 
-// chunked partial sums (grid.y chunks per channel; slab layout [2][chunks][C],
-// every slot plainly written — no zero-init, no atomics on the data path)
-// fused with the finalize reduce via the last-block-done pattern: thread 0 of
-// each block publishes its partials, fences, bumps a per-channel counter, and
-// the block that observes counter==chunks-1 reduces the slab for its channel,
-// folds invstd and the running-stat update, then resets the counter.  One
-// launch instead of two (the finalize launch was ~4 µs of pure launch latency
-// per BN call; ~14 BN fwd/step in the VGG16 bench).
+// chunked partial sums (grid.y chunks per channel -> float atomics) followed
+// by a finalize kernel that also folds invstd and the running-stat update
+// in-kernel (the previous host-side rsqrt/mul_/add_ chain was 5 extra
+// kernel launches per BN call).
+// slab layout [2][chunks][C]: every slot plainly written (no zero-init, no
+// atomics); the finalize kernel reduces over chunks
 __global__ void bn_partial_kernel(const float* __restrict__ x,
                                   float* __restrict__ slab,
-                                  int* __restrict__ counters,
-                                  float* __restrict__ mean,
-                                  float* __restrict__ invstd,
-                                  float* __restrict__ running_mean,
-                                  float* __restrict__ running_var,
-                                  int B, int C, int HW, float momentum,
-                                  float eps) {
+                                  int B, int C, int HW) {
   __shared__ double scratch[16];
   const int c = blockIdx.x;
   const int total = B * HW;
@@ -72,21 +49,28 @@ __global__ void bn_partial_kernel(const float* __restrict__ x,
   double ts = slk_block_sum(s, scratch);
   __syncthreads();
   double ts2 = slk_block_sum(s2, scratch);
-  if (threadIdx.x != 0) return;
-  const int chunks = gridDim.y;
-  slab[(long)blockIdx.y * C + c] = (float)ts;
-  slab[(long)chunks * C + (long)blockIdx.y * C + c] = (float)ts2;
-  __threadfence();
-  if (atomicAdd(&counters[c], 1) != chunks - 1) return;
-  __threadfence();  // order the slab reads after the observed atomic
-  float fs = 0.f, fs2 = 0.f;
-  for (int k = 0; k < chunks; ++k) {
-    fs += slab[(long)k * C + c];
-    fs2 += slab[(long)chunks * C + (long)k * C + c];
+  if (threadIdx.x == 0) {
+    const long chunks = gridDim.y;
+    slab[(long)blockIdx.y * C + c] = (float)ts;
+    slab[chunks * C + (long)blockIdx.y * C + c] = (float)ts2;
   }
-  const float n = (float)total;
-  const float m = fs / n;
-  const float v = fmaxf(fs2 / n - m * m, 0.f);
+}
+
+__global__ void bn_finalize_kernel(const float* __restrict__ slab, int chunks,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ invstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var, int C, float n,
+                                   float momentum, float eps) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, s2 = 0.f;
+  for (int k = 0; k < chunks; ++k) {
+    s += slab[(long)k * C + c];
+    s2 += slab[(long)chunks * C + (long)k * C + c];
+  }
+  const float m = s / n;
+  const float v = fmaxf(s2 / n - m * m, 0.f);
   mean[c] = m;
   invstd[c] = rsqrtf(v + eps);
   if (running_mean != nullptr) {
@@ -94,7 +78,6 @@ __global__ void bn_partial_kernel(const float* __restrict__ x,
     running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
     running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
   }
-  counters[c] = 0;  // kernel-exit is a full fence: next launch sees 0
 }
 
 __global__ void bn_fwd_kernel(const float* __restrict__ x, float* __restrict__ y,
@@ -114,16 +97,13 @@ __global__ void bn_fwd_kernel(const float* __restrict__ x, float* __restrict__ y
 
 // reductions for backward: sum(gy) and sum(gy * xhat) per channel, chunked
 // over grid.y like the forward stats (one block per channel leaves most CUs
-// idle at C=64); finalize fused in via the same last-block-done pattern.
+// idle at C=64)
 __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
                                      const float* __restrict__ gy,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
-                                     float* __restrict__ slab,
-                                     int* __restrict__ counters,
-                                     float* __restrict__ sum_gy,
-                                     float* __restrict__ sum_gy_xhat,
-                                     int B, int C, int HW) {
+                                     float* __restrict__ slab, int B, int C,
+                                     int HW) {
   __shared__ double scratch[16];
   const int c = blockIdx.x;
   const float m = mean[c], is = invstd[c];
@@ -143,21 +123,25 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
   double ts = slk_block_sum(s, scratch);
   __syncthreads();
   double tsx = slk_block_sum(sx, scratch);
-  if (threadIdx.x != 0) return;
-  const int chunks = gridDim.y;
-  slab[(long)blockIdx.y * C + c] = (float)ts;
-  slab[(long)chunks * C + (long)blockIdx.y * C + c] = (float)tsx;
-  __threadfence();
-  if (atomicAdd(&counters[c], 1) != chunks - 1) return;
-  __threadfence();
-  float fs = 0.f, fsx = 0.f;
-  for (int k = 0; k < chunks; ++k) {
-    fs += slab[(long)k * C + c];
-    fsx += slab[(long)chunks * C + (long)k * C + c];
+  if (threadIdx.x == 0) {
+    const long chunks = gridDim.y;
+    slab[(long)blockIdx.y * C + c] = (float)ts;
+    slab[chunks * C + (long)blockIdx.y * C + c] = (float)tsx;
   }
-  sum_gy[c] = fs;
-  sum_gy_xhat[c] = fsx;
-  counters[c] = 0;
+}
+
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ slab, int chunks,
+                                       float* __restrict__ sum_gy,
+                                       float* __restrict__ sum_gy_xhat, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, sx = 0.f;
+  for (int k = 0; k < chunks; ++k) {
+    s += slab[(long)k * C + c];
+    sx += slab[(long)chunks * C + (long)k * C + c];
+  }
+  sum_gy[c] = s;
+  sum_gy_xhat[c] = sx;
 }
 
 __global__ void bn_bwd_dx_kernel(const float* __restrict__ x,
@@ -201,14 +185,15 @@ std::vector<at::Tensor> bn2d_stats_fused(const at::Tensor& x,
   const int chunks = bn_chunks((long)B * HW);
   auto slab = at::empty({2, chunks, C}, x.options());
   hipLaunchKernelGGL(bn_partial_kernel, dim3(C, chunks), dim3(256), 0, stream,
-                     x.data_ptr<float>(), slab.data_ptr<float>(),
-                     slk_counter_buf(C, stream),
+                     x.data_ptr<float>(), slab.data_ptr<float>(), B, C, HW);
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
+                     stream, slab.data_ptr<float>(), chunks,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      running_mean.has_value() ? running_mean->data_ptr<float>()
                                               : nullptr,
                      running_var.has_value() ? running_var->data_ptr<float>()
                                              : nullptr,
-                     B, C, HW, (float)momentum, (float)eps);
+                     C, (float)((long)B * HW), (float)momentum, (float)eps);
   return {mean, invstd};
 }
 
@@ -244,9 +229,10 @@ static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tens
                      dim3(256), 0, stream,
                      x.data_ptr<float>(), gy.data_ptr<float>(),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     slab.data_ptr<float>(), slk_counter_buf(C, stream),
-                     sum_gy.data_ptr<float>(), sum_gy_xhat.data_ptr<float>(),
-                     B, C, HW);
+                     slab.data_ptr<float>(), B, C, HW);
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
+                     stream, slab.data_ptr<float>(), rchunks,
+                     sum_gy.data_ptr<float>(), sum_gy_xhat.data_ptr<float>(), C);
   int grid = (int)std::min<long>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(grid), dim3(256), 0, stream,
                      x.data_ptr<float>(), gy.data_ptr<float>(),
