@@ -52,6 +52,8 @@ def _flags():
         "-Wno-deprecated-declarations",
         "-fno-gpu-rdc",
     ] + [f"-I{p}" for p in includes]
+    if os.environ.get("SLK_MFMA32"):  # 32x32x2 MFMA core variant (A/B experiments)
+        cxxflags.append("-DSLK_MFMA32")
     ldflags = [
         "-shared",
         f"-L{torch_lib}", "-ltorch", "-ltorch_cpu", "-ltorch_python", "-lc10",

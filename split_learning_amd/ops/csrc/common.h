@@ -8,7 +8,9 @@
 
 #define SLK_WAVE 64
 
+using f32x2 = __attribute__((ext_vector_type(2))) float;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
+using f32x16 = __attribute__((ext_vector_type(16))) float;
 
 #define HIP_CHECK(expr)                                                          \
   do {                                                                           \
@@ -27,13 +29,17 @@ static inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
 // bisection — see profiles/SUMMARY.md "graph divergence").  Kernel nodes
 // order correctly; cost is the same single launch.
 static __global__ void slk_zero_kernel(float* __restrict__ p, long n) {
+  // float4 main body + scalar tail (torch allocations are 16B-aligned)
+  const long n4 = n >> 2;
+  f32x4* __restrict__ p4 = reinterpret_cast<f32x4*>(p);
   const long stride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
-    p[i] = 0.f;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (long i = i0; i < n4; i += stride) p4[i] = f32x4{};
+  for (long i = (n4 << 2) + i0; i < n; i += stride) p[i] = 0.f;
 }
 
 static inline void slk_zero_async(float* p, long n, hipStream_t stream) {
-  const int grid = (int)std::min<long>((n + 255) / 256, 2048);
+  const int grid = (int)std::min<long>((n / 4 + 255) / 256 + 1, 4096);
   hipLaunchKernelGGL(slk_zero_kernel, dim3(grid), dim3(256), 0, stream, p, n);
 }
 

@@ -28,13 +28,37 @@
 constexpr int SLK_BM = 64;
 constexpr int SLK_BN = 64;
 constexpr int SLK_BK = 16;
-constexpr int SLK_LDS_PAD = 4;
+
+// LDS layout for the staged A/B tiles: PMC analysis (profiles/SUMMARY.md "LDS
+// conflict fix") showed the per-CU LDS array saturated — occupancy is maxed
+// (36-40 VGPR -> 8 waves/SIMD) yet MFMA sits at 28% busy with
+// SQ_LDS_BANK_CONFLICT ~400/wave: the old [k][m] rows at stride BM+4 (≡ 4 mod
+// 32 banks) put the two 16-lane row-groups of every ds_read_b32 on 12
+// overlapping banks (2-way), and operand pairs (m, m+16) were not adjacent so
+// each needed its own b32 read.
+//
+// phys(k, m) = k*96 + 32*(m>>5) + ((2*(m&15) + ((m>>4)&1)) ^ (2*k)):
+//   * pair-adjacency: cols m and m+16 differ only in (m>>4)&1 -> adjacent
+//     dwords -> ONE ds_read_b64 per operand pair (8 wide reads per tile
+//     instead of 16 narrow: LDS is 64 dwords/clk for b64, 32 for b32);
+//   * row separation: stride 96 ≡ 32 (mod 64 b64-banks) + the 32*(m>>5) half
+//     split puts the two rows of a read's lane groups in disjoint bank
+//     halves -> conflict-free reads;
+//   * the ^(2*k) swizzle spreads the k-fast staging writes (16 rows x 2 cols
+//     per 32-lane group) over 16 banks -> writes stay 2-way (same as the old
+//     layout), and it preserves pair adjacency (even, low-5-bit operand).
+constexpr int SLK_LDS_ROW = 96;
+
+__device__ __forceinline__ int slk_lds_phys(int k, int m) {
+  return k * SLK_LDS_ROW + 32 * (m >> 5) +
+         (((2 * (m & 15)) | ((m >> 4) & 1)) ^ (2 * k));
+}
 
 template <typename Gather, typename Store>
 __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     Gather g, Store st, int M, int N, int K, int split_k, int k_per_split) {
-  __shared__ float ldsA[SLK_BK][SLK_BM + SLK_LDS_PAD];
-  __shared__ float ldsB[SLK_BK][SLK_BN + SLK_LDS_PAD];
+  __shared__ __align__(16) float ldsA[SLK_BK * SLK_LDS_ROW];
+  __shared__ __align__(16) float ldsB[SLK_BK * SLK_LDS_ROW];
 
   const int tile_n = blockIdx.x;
   const int tile_m = blockIdx.y;
@@ -55,7 +79,16 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
   const int frag_r = lane >> 4;    // 0..3: k sub-index for A/B operands
   const int frag_c = lane & 15;    // 0..15
 
+#ifdef SLK_MFMA32
+  // 32x32x2 variant: one accumulator chain per wave (issue interval ==
+  // dependent latency == 64 cyc, so a single chain back-to-backs), 8 MFMA
+  // instructions per BK=16 tile instead of 16 — fewer issue slots in an
+  // issue-bound kernel; k-order (and thus numerics) bit-identical.
+  f32x16 acc32 = {};
+  (void)frag_r; (void)frag_c;
+#else
   f32x4 acc[2][2] = {};
+#endif
 
   constexpr int RA = (SLK_BM * SLK_BK) / 256;  // per-thread A elements
   constexpr int RB = (SLK_BN * SLK_BK) / 256;  // per-thread B elements
@@ -103,31 +136,55 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     #pragma unroll
     for (int i = 0; i < RA; ++i) {
       const int idx = tid + i * 256;
-      ldsA[idx & 15][idx >> 4] = ra[i];
+      ldsA[slk_lds_phys(idx & 15, idx >> 4)] = ra[i];
     }
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
       const int idx = tid + i * 256;
-      ldsB[idx >> 6][idx & 63] = rb[i];
+      ldsB[slk_lds_phys(idx >> 6, idx & 63)] = rb[i];
     }
     __syncthreads();
     if (k0 + SLK_BK < k_end) load_tile(k0 + SLK_BK);  // overlaps the MFMAs
 
+#ifdef SLK_MFMA32
+    // operand map (ISA): lane l supplies A[i=l&31][k=l>>5], B[k=l>>5][j=l&31]
+    const int kh = lane >> 5;
+    const int l31 = lane & 31;
+    #pragma unroll
+    for (int kk = 0; kk < SLK_BK / 2; ++kk) {
+      float a = ldsA[slk_lds_phys(kk * 2 + kh, wm + l31)];
+      float b = ldsB[slk_lds_phys(kk * 2 + kh, wn + l31)];
+      acc32 = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc32, 0, 0, 0);
+    }
+#else
+    // operand pairs (frag_c, frag_c+16) are adjacent dwords in the swizzled
+    // layout -> each f32x2 load is one conflict-free ds_read_b64
     #pragma unroll
     for (int kk = 0; kk < SLK_BK / 4; ++kk) {
       const int kr = kk * 4 + frag_r;
-      float a0 = ldsA[kr][wm + frag_c];
-      float a1 = ldsA[kr][wm + 16 + frag_c];
-      float b0 = ldsB[kr][wn + frag_c];
-      float b1 = ldsB[kr][wn + 16 + frag_c];
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+      f32x2 a01 = *reinterpret_cast<const f32x2*>(
+          &ldsA[slk_lds_phys(kr, wm + frag_c)]);
+      f32x2 b01 = *reinterpret_cast<const f32x2*>(
+          &ldsB[slk_lds_phys(kr, wn + frag_c)]);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.x, b01.x, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.x, b01.y, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.y, b01.x, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.y, b01.y, acc[1][1], 0, 0, 0);
     }
+#endif
     __syncthreads();
   }
 
+#ifdef SLK_MFMA32
+  // epilogue: C/D map for 32x32 shapes: col = lane&31,
+  // row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  #pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int row = m0 + wm + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+    int col = n0 + wn + (lane & 31);
+    if (row < M && col < N) st.store(batch, row, col, acc32[r], ks == 0);
+  }
+#else
   // epilogue: C/D fragment mapping for 16x16x4: col = lane&15, row = (lane>>4)*4 + i
   #pragma unroll
   for (int mi = 0; mi < 2; ++mi) {
@@ -141,6 +198,7 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
       }
     }
   }
+#endif
 }
 
 // Host-side launch helper.
