@@ -24,10 +24,48 @@ from . import native
 # GEMM / Linear
 # ---------------------------------------------------------------------------
 
+def _matmul_raw(a, b, trans_a=False, trans_b=False, out=None, accumulate=False):
+    """Non-differentiable strided MFMA GEMM (internal / backward use)."""
+    return native().matmul_f32(a, b, trans_a, trans_b, out, accumulate)
+
+
+class MatmulFn(Function):
+    """Autograd for C = op(A) @ op(B): without this, the raw pybind op returns
+    a detached tensor and gradient flow through the attention score/context
+    matmuls silently stops (in_proj and the q/k/v projections would never
+    train on GPU while the CPU fallback path trains — caught by
+    tests/test_kernels_gpu.py::test_attention_core_grads)."""
+
+    @staticmethod
+    def forward(ctx, a, b, trans_a, trans_b):
+        ctx.save_for_backward(a, b)
+        ctx.trans = (trans_a, trans_b)
+        return native().matmul_f32(a, b, trans_a, trans_b, None, False)
+
+    @staticmethod
+    def backward(ctx, gy):
+        a, b = ctx.saved_tensors
+        ta, tb = ctx.trans
+        gy = gy.contiguous()
+        ga = gb = None
+        if ctx.needs_input_grad[0]:
+            ga = (_matmul_raw(gy, b, trans_a=False, trans_b=not tb) if not ta
+                  else _matmul_raw(b, gy, trans_a=tb, trans_b=True))
+        if ctx.needs_input_grad[1]:
+            gb = (_matmul_raw(a, gy, trans_a=not ta, trans_b=False) if not tb
+                  else _matmul_raw(gy, a, trans_a=True, trans_b=ta))
+        return ga, gb, None, None
+
+
 def matmul_f32(a: torch.Tensor, b: torch.Tensor, trans_a: bool = False, trans_b: bool = False,
                out: Optional[torch.Tensor] = None, accumulate: bool = False) -> torch.Tensor:
-    """C = op(A) @ op(B) for 2-D or batched 3-D fp32 tensors (strided MFMA GEMM)."""
-    return native().matmul_f32(a, b, trans_a, trans_b, out, accumulate)
+    """C = op(A) @ op(B) for 2-D or batched 3-D fp32 tensors (strided MFMA GEMM).
+
+    Differentiable unless `out`/`accumulate` are used (those forms are
+    internal building blocks and bypass autograd)."""
+    if out is not None or accumulate:
+        return native().matmul_f32(a, b, trans_a, trans_b, out, accumulate)
+    return MatmulFn.apply(a, b, trans_a, trans_b)
 
 
 class LinearFn(Function):
@@ -47,10 +85,10 @@ class LinearFn(Function):
         gys = gy.reshape(-1, gy.shape[-1]).contiguous()
         gx = gw = gb = None
         if ctx.needs_input_grad[0]:
-            gx = matmul_f32(gys, weight)               # [M,N]x[N,K] -> [M,K]
+            gx = _matmul_raw(gys, weight)              # [M,N]x[N,K] -> [M,K]
             gx = gx.reshape(ctx.x_shape)
         if ctx.needs_input_grad[1]:
-            gw = matmul_f32(gys, xs, trans_a=True)     # [N,M]x[M,K] -> [N,K]
+            gw = _matmul_raw(gys, xs, trans_a=True)    # [N,M]x[M,K] -> [N,K]
         if ctx.has_bias and ctx.needs_input_grad[2]:
             gb = native().colsum_f32(gys)
         return gx, gw, gb
