@@ -12,12 +12,31 @@ the in-tree _sl_kernels extension (see ops/__init__.py loading policy).
 
 from __future__ import annotations
 
+import os
 from typing import Optional, Tuple
 
 import torch
+import torch.nn.functional as _F
 from torch.autograd import Function
 
 from . import native
+
+# Large plain GEMMs route to the platform BLAS (rocBLAS/hipBLASLt via
+# torch.matmul / F.linear): measured per-model on MI355X (profiles/SUMMARY.md
+# "library-GEMM routing"), our 64x64-tile MFMA kernel wins for the small/skinny
+# shapes of the conv nets (VGG16 linear: 11.8k vs 10.9k samples/s) while the
+# library wins ~2x for transformer shapes (ViT 3.2k vs 1.6k, KWT 1.2k vs
+# 0.57k).  Thresholds are in FLOPs of the whole call, set by a GPU sweep
+# (MM 1e6..5e7 x LIN 1e7..2e8); env-overridable for sweeps.  Fused / gather /
+# split-K ops (conv, BN, optimizers, attention softmax) always stay on the
+# hand-written kernels.
+_LIB_MM_THRESH = float(os.environ.get("SLK_LIB_MM_THRESH", "1e7"))
+_LIB_LIN_THRESH = float(os.environ.get("SLK_LIB_LIN_THRESH", "2e7"))
+
+
+def use_lib_mm(batch, m, n, k) -> bool:
+    """True when a batched matmul of this size routes to the platform BLAS."""
+    return 2.0 * batch * m * n * k >= _LIB_MM_THRESH
 
 
 # ---------------------------------------------------------------------------
@@ -38,6 +57,8 @@ class MatmulFn(Function):
 
     @staticmethod
     def forward(ctx, a, b, trans_a, trans_b):
+        a = a.contiguous()
+        b = b.contiguous()
         ctx.save_for_backward(a, b)
         ctx.trans = (trans_a, trans_b)
         return native().matmul_f32(a, b, trans_a, trans_b, None, False)
@@ -65,6 +86,14 @@ def matmul_f32(a: torch.Tensor, b: torch.Tensor, trans_a: bool = False, trans_b:
     internal building blocks and bypass autograd)."""
     if out is not None or accumulate:
         return native().matmul_f32(a, b, trans_a, trans_b, out, accumulate)
+    m = a.shape[-1] if trans_a else a.shape[-2]
+    k = a.shape[-2] if trans_a else a.shape[-1]
+    n = b.shape[-2] if trans_b else b.shape[-1]
+    batch = a.shape[0] if a.dim() == 3 else 1
+    if 2.0 * batch * m * n * k >= _LIB_MM_THRESH:
+        av = a.transpose(-1, -2) if trans_a else a
+        bv = b.transpose(-1, -2) if trans_b else b
+        return torch.matmul(av, bv)
     return MatmulFn.apply(a, b, trans_a, trans_b)
 
 
@@ -95,6 +124,9 @@ class LinearFn(Function):
 
 
 def linear(x, weight, bias=None):
+    rows = x.numel() // x.shape[-1]
+    if 2.0 * rows * weight.shape[0] * weight.shape[1] >= _LIB_LIN_THRESH:
+        return _F.linear(x, weight, bias)
     return LinearFn.apply(x, weight, bias)
 
 

@@ -123,7 +123,8 @@ class HipMultiheadAttention(nn.MultiheadAttention):
     """
 
     def forward(self, query, key, value, **kwargs):  # type: ignore[override]
-        if not _on_gpu(query) or (query is not key) or (key is not value) or not self.batch_first:
+        if (not _use_native("attn", query) or (query is not key)
+                or (key is not value) or not self.batch_first):
             return super().forward(query, key, value, **kwargs)
         x = query  # [B, S, E]
         B, S, E = x.shape
@@ -131,16 +132,31 @@ class HipMultiheadAttention(nn.MultiheadAttention):
         hd = E // H
         qkv = hf.linear(x, self.in_proj_weight, self.in_proj_bias)  # [B,S,3E]
         q, k, v = qkv.split(E, dim=-1)
-        # [B,S,H,hd] -> [B*H, S, hd]
-        q = q.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd).contiguous()
-        k = k.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd).contiguous()
-        v = v.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd).contiguous()
-        scores = hf.matmul_f32(q, k, trans_b=True) * (1.0 / math.sqrt(hd))
-        probs = hf.softmax_lastdim(scores)
-        if self.dropout > 0.0:
-            probs = hf.dropout(probs, self.dropout, self.training)
-        ctxv = hf.matmul_f32(probs, v)  # [B*H, S, hd]
-        ctxv = ctxv.reshape(B, H, S, hd).permute(0, 2, 1, 3).reshape(B, S, E)
+        scale = 1.0 / math.sqrt(hd)
+        if hf.use_lib_mm(B * H, S, S, hd):
+            # library route (rocBLAS batched GEMM): keep [B,H,S,hd] strided
+            # VIEWS end-to-end — no contiguous copies; softmax/dropout stay
+            # on the fused HIP kernels
+            q = q.reshape(B, S, H, hd).transpose(1, 2)
+            k = k.reshape(B, S, H, hd).transpose(1, 2)
+            v = v.reshape(B, S, H, hd).transpose(1, 2)
+            scores = torch.matmul(q, k.transpose(-1, -2)) * scale
+            probs = hf.softmax_lastdim(scores)
+            if self.dropout > 0.0:
+                probs = hf.dropout(probs, self.dropout, self.training)
+            ctxv = torch.matmul(probs, v)  # [B, H, S, hd]
+            ctxv = ctxv.transpose(1, 2).reshape(B, S, E)
+        else:
+            # native MFMA route: [B*H, S, hd] contiguous batches
+            q = q.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd)
+            k = k.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd)
+            v = v.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd)
+            scores = hf.matmul_f32(q, k, trans_b=True) * scale
+            probs = hf.softmax_lastdim(scores)
+            if self.dropout > 0.0:
+                probs = hf.dropout(probs, self.dropout, self.training)
+            ctxv = hf.matmul_f32(probs, v)  # [B*H, S, hd]
+            ctxv = ctxv.reshape(B, H, S, hd).permute(0, 2, 1, 3).reshape(B, S, E)
         out = hf.linear(ctxv, self.out_proj.weight, self.out_proj.bias)
         return out, None
 
@@ -150,7 +166,7 @@ def attention_core(q, k, v, dropout_p: float = 0.0, training: bool = False,
     """scaled-dot-product attention on [B*H, S, hd] tensors via HIP kernels."""
     hd = q.shape[-1]
     s = scale if scale is not None else 1.0 / math.sqrt(hd)
-    if q.is_cuda:
+    if _use_native("attn", q):
         scores = hf.matmul_f32(q, k, trans_b=True) * s
         probs = hf.softmax_lastdim(scores)
         if dropout_p > 0.0:
