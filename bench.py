@@ -294,9 +294,9 @@ def main():
     ap.add_argument("--warmup", type=int, default=16)
     ap.add_argument("--device", default=None)
     ap.add_argument("--graphs", action="store_true",
-                    help="hipGraph capture of the serial N=1 step (the "
-                         "default N=1 mode is the faster two-stream 1F1B "
-                         "overlap; see profiles/SUMMARY.md)")
+                    help="hipGraph capture of the serial N=1 step (measured "
+                         "tied with the plain serial default; see "
+                         "profiles/SUMMARY.md)")
     ap.add_argument("--no-graphs", action="store_true",
                     help="(compat) force-disable graph capture")
     ap.add_argument("--serial", action="store_true",
