@@ -25,6 +25,7 @@ SOURCES = [
     "norm.hip",
     "elementwise.hip",
     "loss.hip",
+    "attention.hip",
     "optim.hip",
     "bindings.cpp",
 ]

@@ -45,6 +45,9 @@ std::vector<at::Tensor> maxpool2x2_fwd(const at::Tensor&);
 at::Tensor maxpool2x2_bwd(const at::Tensor&, const at::Tensor&, int, int);
 at::Tensor embedding_fwd(const at::Tensor&, const at::Tensor&);
 at::Tensor embedding_bwd(const at::Tensor&, const at::Tensor&, int64_t, int64_t);
+// attention.hip
+std::vector<at::Tensor> attn_fwd(const at::Tensor&, const at::Tensor&,
+                                 const at::Tensor&, double);
 // loss.hip
 at::Tensor softmax_fwd(const at::Tensor&);
 at::Tensor softmax_bwd(const at::Tensor&, const at::Tensor&);
@@ -94,6 +97,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2x2_bwd", &slk::maxpool2x2_bwd);
   m.def("embedding_fwd", &slk::embedding_fwd);
   m.def("embedding_bwd", &slk::embedding_bwd);
+  m.def("attn_fwd", &slk::attn_fwd);
   m.def("softmax_fwd", &slk::softmax_fwd);
   m.def("softmax_bwd", &slk::softmax_bwd);
   m.def("ce_fwd", &slk::ce_fwd);

@@ -359,6 +359,52 @@ def layer_norm(x, gamma, beta, eps=1e-5):
 
 
 # ---------------------------------------------------------------------------
+# Fused attention (small-S SDPA: S<=128, hd in {32,64} — the model zoo's regime)
+# ---------------------------------------------------------------------------
+
+class AttnFn(Function):
+    """softmax(Q@K^T * scale) @ V in ONE forward kernel (attention.hip): K/V
+    live in LDS, no scores round-trip through HBM, no separate scale/softmax
+    launches.  Backward reuses the GEMM + softmax-bwd kernels on the saved
+    probability matrix (identical math to the unfused path)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        out, probs = native().attn_fwd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, probs)
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, gy):
+        q, k, v, probs = ctx.saved_tensors
+        gy = gy.contiguous()
+        # dV = P^T @ gO ; dP = gO @ V^T ; dS = softmax_bwd(dP) * scale
+        gv = _matmul_raw(probs, gy, trans_a=True)
+        gp = _matmul_raw(gy, v, trans_b=True)
+        bhs = probs.shape[0] * probs.shape[1]
+        gs = native().softmax_bwd(gp.reshape(bhs, -1).contiguous(),
+                                  probs.reshape(bhs, -1)).reshape(probs.shape)
+        gs = gs * ctx.scale
+        gq = _matmul_raw(gs, k)
+        gk = _matmul_raw(gs, q, trans_a=True)
+        return gq, gk, gv, None
+
+
+def attn_fused_ok(q) -> bool:
+    """fused path precondition: [BH, S<=128, hd<=64 pow2] fp32 CUDA."""
+    if not (q.is_cuda and q.dim() == 3 and q.dtype == torch.float32):
+        return False
+    s, hd = q.shape[1], q.shape[2]
+    return s <= 128 and hd <= 64 and (hd & (hd - 1)) == 0
+
+
+def attention(q, k, v, scale):
+    """Fused SDPA (no dropout variant); inputs [BH, S, hd] contiguous."""
+    return AttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(), scale)
+
+
+# ---------------------------------------------------------------------------
 # Softmax (last dim) — used by attention
 # ---------------------------------------------------------------------------
 

@@ -93,6 +93,28 @@ def test_attention_core_grads():
         assert_close(ours, theirs, atol=1e-3, rtol=1e-3, what=f"attention {name}")
 
 
+@pytest.mark.parametrize("s,hd", [(65, 32), (99, 64), (128, 64), (7, 64)])
+def test_attn_fused_fwd_bwd(s, hd):
+    """Fused SDPA kernel (attention.hip) vs torch reference, fwd + all grads."""
+    from split_learning_amd.ops import functional as hf
+    torch.manual_seed(2)
+    q = torch.randn(12, s, hd, device="cuda", requires_grad=True)
+    k = torch.randn(12, s, hd, device="cuda", requires_grad=True)
+    v = torch.randn(12, s, hd, device="cuda", requires_grad=True)
+    qr, kr, vr = (t.detach().clone().requires_grad_(True) for t in (q, k, v))
+    scale = 1.0 / (hd ** 0.5)
+    out = hf.attention(q, k, v, scale)
+    ref = torch.matmul(F.softmax(torch.matmul(qr, kr.transpose(-1, -2)) * scale,
+                                 dim=-1), vr)
+    assert_close(out, ref, atol=1e-3, rtol=1e-3, what=f"attn_fwd s={s} hd={hd}")
+    g = torch.randn_like(out)
+    out.backward(g)
+    ref.backward(g)
+    assert_close(q.grad, qr.grad, atol=1e-3, rtol=1e-3, what="attn gq")
+    assert_close(k.grad, kr.grad, atol=1e-3, rtol=1e-3, what="attn gk")
+    assert_close(v.grad, vr.grad, atol=1e-3, rtol=1e-3, what="attn gv")
+
+
 def test_mha_module_grads():
     """HipMultiheadAttention vs nn.MultiheadAttention with identical weights:
     forward parity and in_proj gradient flow."""
