@@ -284,7 +284,12 @@ class DistPipeline:
                 self.opt.step()
                 self.plane.send_gradient(1, self.peer, GradientMsg(
                     m.data_id, act.grad.detach(), []))
-        return nan_flag
+        # the loss (and thus the NaN flag) lives on stage-2 ranks; surface it
+        # on rank 0 where the JSON line is printed
+        import torch.distributed as dist
+        f = nan_flag.to(torch.int32)
+        dist.all_reduce(f, op=dist.ReduceOp.MAX)
+        return f.bool()
 
 
 def main():
