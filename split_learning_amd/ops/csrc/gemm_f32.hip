@@ -89,7 +89,7 @@ static void launch_strided(const at::Tensor& a2, const at::Tensor& b2, at::Tenso
   st.accumulate = accumulate || split_k > 1;
 
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  slk_launch_gemm(g, st, M, N, K, nb, split_k, stream);
+  slk_launch_gemm(g, st, M, N, K, nb, split_k, stream, slk_use_big(M, N));
 }
 
 at::Tensor matmul_f32(const at::Tensor& a, const at::Tensor& b, bool ta, bool tb,
@@ -115,7 +115,9 @@ at::Tensor matmul_f32(const at::Tensor& a, const at::Tensor& b, bool ta, bool tb
   }
   int split_k = 1;
   if (!accumulate && !out.has_value()) {
-    split_k = slk_pick_split_k(M, N, Ka, a.dim() == 3 ? a.size(0) : 1);
+    const int bm = slk_use_big(M, N) ? SLK_BM2 : SLK_BM;
+    const int bn = slk_use_big(M, N) ? SLK_BN2 : SLK_BN;
+    split_k = slk_pick_split_k(M, N, Ka, a.dim() == 3 ? a.size(0) : 1, bm, bn);
     if (split_k > 1) {
       slk_zero_async(c.data_ptr<float>(), c.numel(),
                      c10::hip::getCurrentHIPStream().stream());
@@ -136,7 +138,9 @@ at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
   TORCH_CHECK(!bias.has_value() || bias->is_cuda(), "linear_fwd: bias device");
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1));
   const int M = x.size(0), N = w.size(0), K = x.size(1);
-  int split_k = slk_pick_split_k(M, N, K, 1);
+  const bool big = slk_use_big(M, N);
+  int split_k = slk_pick_split_k(M, N, K, 1, big ? SLK_BM2 : SLK_BM,
+                                 big ? SLK_BN2 : SLK_BN);
   auto y = split_k > 1 ? zeroed({M, N}, x.options())
                        : at::empty({M, N}, x.options());
   launch_strided(x, w, y, /*ta=*/false, /*tb=*/true, bias, /*acc=*/false, split_k);

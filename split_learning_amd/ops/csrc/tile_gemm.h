@@ -201,15 +201,155 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
 #endif
 }
 
+// ---------------------------------------------------------------------------
+// Big-tile variant: 128x128 block, 64x64 per wave as a 2x2 grid of
+// v_mfma_f32_32x32x2_f32 chains (16 accumulator VGPRs each, issue interval ==
+// dependent latency == 64 cyc so the 4 interleaved chains saturate the MAI
+// pipe).  4x the MFMA work per staged element vs the 64x64 kernel — built to
+// close the large plain-GEMM gap vs rocBLAS (ROADMAP.md item 2; the 64x64
+// kernel measures 91 TF at 4096^3 vs rocBLAS 150).  Experimental: selected by
+// SLK_GEMM_BIG=1 for large strided GEMMs only (gemm_f32.hip), never conv.
+//
+// LDS stride 130 (== 2 mod 32 banks): staging writes (16 k-rows x 2 cols per
+// 32-lane group) land on 32 distinct banks (2k+m); mfma32 operand reads are
+// whole-row (all 32 lanes of a group share k), hence conflict-free at any
+// stride.
+constexpr int SLK_BM2 = 128;
+constexpr int SLK_BN2 = 128;
+constexpr int SLK_LDS_ROW2 = 130;
+
+template <typename Gather, typename Store>
+__global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
+    Gather g, Store st, int M, int N, int K, int split_k, int k_per_split) {
+  __shared__ float ldsA[SLK_BK * SLK_LDS_ROW2];
+  __shared__ float ldsB[SLK_BK * SLK_LDS_ROW2];
+
+  const int tile_n = blockIdx.x;
+  const int tile_m = blockIdx.y;
+  const int batch = blockIdx.z / split_k;
+  const int ks = blockIdx.z % split_k;
+
+  const int m0 = tile_m * SLK_BM2;
+  const int n0 = tile_n * SLK_BN2;
+  const int k_begin = ks * k_per_split;
+  const int k_end = min(K, k_begin + k_per_split);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * 64;
+  const int wn = (wid & 1) * 64;
+  const int kh = lane >> 5;      // mfma32 k half: A[i=l&31][k=l>>5]
+  const int l31 = lane & 31;
+
+  f32x16 acc[2][2] = {};         // (mi, ni) 32x32 chains
+
+  constexpr int RA = (SLK_BM2 * SLK_BK) / 256;  // 8
+  constexpr int RB = (SLK_BN2 * SLK_BK) / 256;  // 8
+  float ra[RA], rb[RB];
+
+  typename Gather::ACtx actx[RA];
+  typename Gather::BCtx bctx[RB];
+  int ka[RA], kb[RB];
+  #pragma unroll
+  for (int i = 0; i < RA; ++i) {
+    const int idx = tid + i * 256;
+    const int m = m0 + (idx >> 4);      // k-fast staging, 128 rows
+    ka[i] = idx & 15;
+    actx[i] = g.prepA(batch, min(m, M - 1), m < M);
+  }
+  #pragma unroll
+  for (int i = 0; i < RB; ++i) {
+    const int idx = tid + i * 256;
+    const int n = n0 + (idx & 127);     // n-fast staging, 128 cols
+    kb[i] = idx >> 7;
+    bctx[i] = g.prepB(batch, min(n, N - 1), n < N);
+  }
+
+  auto load_tile = [&](int k0) {
+    #pragma unroll
+    for (int i = 0; i < RA; ++i) {
+      const int k = k0 + ka[i];
+      ra[i] = g.loadA(actx[i], min(k, K - 1), k < k_end);
+    }
+    #pragma unroll
+    for (int i = 0; i < RB; ++i) {
+      // (tid + i*256) >> 7 is wave-uniform (see the 64x64 kernel's note)
+      const int k = __builtin_amdgcn_readfirstlane(k0 + kb[i]);
+      rb[i] = g.loadB(bctx[i], min(k, K - 1), k < k_end);
+    }
+  };
+
+  load_tile(k_begin);
+  for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
+    #pragma unroll
+    for (int i = 0; i < RA; ++i) {
+      const int idx = tid + i * 256;
+      ldsA[(idx & 15) * SLK_LDS_ROW2 + (idx >> 4)] = ra[i];
+    }
+    #pragma unroll
+    for (int i = 0; i < RB; ++i) {
+      const int idx = tid + i * 256;
+      ldsB[(idx >> 7) * SLK_LDS_ROW2 + (idx & 127)] = rb[i];
+    }
+    __syncthreads();
+    if (k0 + SLK_BK < k_end) load_tile(k0 + SLK_BK);
+
+    #pragma unroll
+    for (int kk = 0; kk < SLK_BK / 2; ++kk) {
+      const int kr = kk * 2 + kh;
+      const float a0 = ldsA[kr * SLK_LDS_ROW2 + wm + l31];
+      const float a1 = ldsA[kr * SLK_LDS_ROW2 + wm + 32 + l31];
+      const float b0 = ldsB[kr * SLK_LDS_ROW2 + wn + l31];
+      const float b1 = ldsB[kr * SLK_LDS_ROW2 + wn + 32 + l31];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // C/D map for 32x32 shapes: col = lane&31, row = (r&3) + 8*(r>>2) + 4*(lane>>5)
+  #pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+    #pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = m0 + wm + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+        int col = n0 + wn + ni * 32 + (lane & 31);
+        if (row < M && col < N) st.store(batch, row, col, acc[mi][ni][r], ks == 0);
+      }
+    }
+  }
+}
+
+// SLK_GEMM_BIG=1 enables the 128x128 kernel for large plain GEMMs
+inline bool slk_use_big(int M, int N) {
+  static int v = [] {
+    const char* e = std::getenv("SLK_GEMM_BIG");
+    return e ? atoi(e) : 0;
+  }();
+  return v != 0 && M >= 96 && N >= 96;
+}
+
 // Host-side launch helper.
 template <typename Gather, typename Store>
 inline void slk_launch_gemm(const Gather& g, const Store& st, int M, int N, int K,
-                            int n_batch, int split_k, hipStream_t stream) {
+                            int n_batch, int split_k, hipStream_t stream,
+                            bool big = false) {
   if (split_k < 1) split_k = 1;
   int k_per_split = ceil_div(K, split_k);
   // round k_per_split up to a BK multiple so every split starts aligned
   k_per_split = ((k_per_split + SLK_BK - 1) / SLK_BK) * SLK_BK;
   split_k = ceil_div(K, k_per_split);
+  if (big) {
+    dim3 grid(ceil_div(N, SLK_BN2), ceil_div(M, SLK_BM2), n_batch * split_k);
+    hipLaunchKernelGGL((slk_mfma_gemm_kernel_big<Gather, Store>), grid, dim3(256),
+                       0, stream, g, st, M, N, K, split_k, k_per_split);
+    return;
+  }
   dim3 grid(ceil_div(N, SLK_BN), ceil_div(M, SLK_BM), n_batch * split_k);
   hipLaunchKernelGGL((slk_mfma_gemm_kernel<Gather, Store>), grid, dim3(256), 0,
                      stream, g, st, M, N, K, split_k, k_per_split);
@@ -219,7 +359,8 @@ inline void slk_launch_gemm(const Gather& g, const Store& st, int M, int N, int 
 // the GPU split-K sweep in profiles/ shows throughput rising monotonically to a
 // ~1280-block target — the gather path needs many resident waves, and 48-way
 // fp32 atomic accumulation costs less than the idle CUs it fills).
-inline int slk_pick_split_k(int M, int N, int K, int n_batch) {
+inline int slk_pick_split_k(int M, int N, int K, int n_batch,
+                            int bm = SLK_BM, int bn = SLK_BN) {
   // env-tunable (host-side, read per call so in-process sweeps work):
   // SLK_SPLIT_TARGET = block-count target, SLK_SPLIT_CAP = max split factor
   static auto readenv = [](const char* n, long d) {
@@ -228,7 +369,7 @@ inline int slk_pick_split_k(int M, int N, int K, int n_batch) {
   };
   const long target = readenv("SLK_SPLIT_TARGET", 1280);
   const long cap = readenv("SLK_SPLIT_CAP", 64);
-  long tiles = (long)ceil_div(M, SLK_BM) * ceil_div(N, SLK_BN) * (n_batch > 0 ? n_batch : 1);
+  long tiles = (long)ceil_div(M, bm) * ceil_div(N, bn) * (n_batch > 0 ? n_batch : 1);
   if (tiles >= target || K <= SLK_BK * 2) return 1;
   long want = (target + tiles - 1) / tiles;
   if (want > cap) want = cap;

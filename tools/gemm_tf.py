@@ -12,7 +12,7 @@ import time
 import torch
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
-from split_learning_amd.ops import functional as hf  # noqa: E402
+from split_learning_amd.ops import native  # noqa: E402
 
 
 def bench(fn, iters=20, warm=5):
@@ -32,9 +32,10 @@ def main():
         a = torch.randn(n, n, device=dev)
         b = torch.randn(n, n, device=dev)
         flops = 2.0 * n * n * n
+        ext = native()  # raw op: bypasses the library-GEMM routing on purpose
 
         def ours():
-            hf.matmul_f32(a, b)
+            ext.matmul_f32(a, b, False, False, None, False)
 
         def lib():
             torch.matmul(a, b)
