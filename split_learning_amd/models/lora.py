@@ -10,12 +10,12 @@ state dict is shipped to the server, so .pth files contain plain weights.
 from __future__ import annotations
 
 import math
-from typing import Iterable, List
+from typing import Iterable
 
 import torch
 import torch.nn as nn
 
-from ..ops.modules import HipDropout, HipLinear
+from ..ops.modules import HipDropout
 
 
 class LoRALinear(nn.Module):

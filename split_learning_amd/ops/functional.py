@@ -13,7 +13,7 @@ the in-tree _sl_kernels extension (see ops/__init__.py loading policy).
 from __future__ import annotations
 
 import os
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.nn.functional as _F

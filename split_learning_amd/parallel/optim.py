@@ -11,7 +11,6 @@ from typing import List
 
 import torch
 
-from ..ops import functional as hf
 
 
 class FusedSGD:
