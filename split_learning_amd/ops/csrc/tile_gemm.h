@@ -210,19 +210,26 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
 // kernel measures 91 TF at 4096^3 vs rocBLAS 150).  Experimental: selected by
 // SLK_GEMM_BIG=1 for large strided GEMMs only (gemm_f32.hip), never conv.
 //
-// LDS stride 130 (== 2 mod 32 banks): staging writes (16 k-rows x 2 cols per
-// 32-lane group) land on 32 distinct banks (2k+m); mfma32 operand reads are
-// whole-row (all 32 lanes of a group share k), hence conflict-free at any
-// stride.
+// LDS layout: columns are stored pair-interleaved so a wave's two mfma32
+// operands (cols c and c+32 for its 64-wide tile half) are ADJACENT dwords —
+// one conflict-free ds_read_b64 per operand pair instead of two b32 reads
+// (lane l31 hits b64 banks {2*l31, 2*l31+1}: the full 64-bank row).  Staging
+// writes become 2-way conflicted (same as the 64x64 kernel's writes) — a
+// measured-good trade, reads outnumber writes 2:1.
 constexpr int SLK_BM2 = 128;
 constexpr int SLK_BN2 = 128;
 constexpr int SLK_LDS_ROW2 = 130;
 
+__device__ __forceinline__ int slk_big_perm(int c) {
+  // (c, c+32) -> adjacent dwords within each 64-column half
+  return 2 * (c & 31) + ((c >> 5) & 1) + 64 * (c >> 6);
+}
+
 template <typename Gather, typename Store>
 __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
     Gather g, Store st, int M, int N, int K, int split_k, int k_per_split) {
-  __shared__ float ldsA[SLK_BK * SLK_LDS_ROW2];
-  __shared__ float ldsB[SLK_BK * SLK_LDS_ROW2];
+  __shared__ __align__(16) float ldsA[SLK_BK * SLK_LDS_ROW2];
+  __shared__ __align__(16) float ldsB[SLK_BK * SLK_LDS_ROW2];
 
   const int tile_n = blockIdx.x;
   const int tile_m = blockIdx.y;
@@ -285,12 +292,12 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
     #pragma unroll
     for (int i = 0; i < RA; ++i) {
       const int idx = tid + i * 256;
-      ldsA[(idx & 15) * SLK_LDS_ROW2 + (idx >> 4)] = ra[i];
+      ldsA[(idx & 15) * SLK_LDS_ROW2 + slk_big_perm(idx >> 4)] = ra[i];
     }
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
       const int idx = tid + i * 256;
-      ldsB[(idx >> 7) * SLK_LDS_ROW2 + (idx & 127)] = rb[i];
+      ldsB[(idx >> 7) * SLK_LDS_ROW2 + slk_big_perm(idx & 127)] = rb[i];
     }
     __syncthreads();
     if (k0 + SLK_BK < k_end) load_tile(k0 + SLK_BK);
@@ -298,14 +305,15 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
     #pragma unroll
     for (int kk = 0; kk < SLK_BK / 2; ++kk) {
       const int kr = kk * 2 + kh;
-      const float a0 = ldsA[kr * SLK_LDS_ROW2 + wm + l31];
-      const float a1 = ldsA[kr * SLK_LDS_ROW2 + wm + 32 + l31];
-      const float b0 = ldsB[kr * SLK_LDS_ROW2 + wn + l31];
-      const float b1 = ldsB[kr * SLK_LDS_ROW2 + wn + 32 + l31];
-      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+      // pairs (wm+l31, wm+32+l31) / (wn+l31, wn+32+l31) are adjacent dwords
+      const f32x2 a01 = *reinterpret_cast<const f32x2*>(
+          &ldsA[kr * SLK_LDS_ROW2 + 64 * (wm >> 6) + 2 * l31]);
+      const f32x2 b01 = *reinterpret_cast<const f32x2*>(
+          &ldsB[kr * SLK_LDS_ROW2 + 64 * (wn >> 6) + 2 * l31]);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.x, b01.x, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.x, b01.y, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.y, b01.x, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.y, b01.y, acc[1][1], 0, 0, 0);
     }
     __syncthreads();
   }
