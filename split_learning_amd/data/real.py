@@ -15,8 +15,10 @@ import torch
 
 DATA_ROOT = os.environ.get("SL_DATA_ROOT", os.path.join(os.getcwd(), "data"))
 
+# reference normalization constants (src/dataset/dataloader.py CIFAR10
+# transform) so checkpoints stay input-distribution-compatible
 _CIFAR_MEAN = np.array([0.4914, 0.4822, 0.4465], dtype=np.float32)
-_CIFAR_STD = np.array([0.2470, 0.2435, 0.2616], dtype=np.float32)
+_CIFAR_STD = np.array([0.2023, 0.1994, 0.2010], dtype=np.float32)
 
 
 def _subset_by_distribution(x: torch.Tensor, y: torch.Tensor,
@@ -130,7 +132,10 @@ def _load_speechcommands(train: bool, max_per_class: int = 400):
 
 
 def load_real(data_name: str, distribution: Optional[List[int]],
-              train: bool) -> Optional[Tuple[torch.Tensor, torch.Tensor]]:
+              train: bool, seed: int = 0) -> Optional[Tuple[torch.Tensor, torch.Tensor]]:
+    """seed: per-client draw so same-distribution clients select DIFFERENT
+    sample subsets (the reference uses unseeded random.sample per client —
+    ADVICE.md round-1 medium finding)."""
     loaders = {"CIFAR10": _load_cifar10, "MNIST": _load_mnist,
                "AGNEWS": _load_agnews, "SPEECHCOMMANDS": _load_speechcommands}
     fn = loaders.get(data_name)
@@ -139,4 +144,4 @@ def load_real(data_name: str, distribution: Optional[List[int]],
     loaded = fn(train)
     if loaded is None:
         return None
-    return _subset_by_distribution(*loaded, distribution)
+    return _subset_by_distribution(*loaded, distribution, seed=seed)

@@ -283,6 +283,9 @@ class P2PData:
         n = len(self.up_peers)
         if n == 0:
             return None
+        import time as _time
+        t0 = _time.monotonic() if (block and timeout) else None
+        empty_sweeps = 0
         while True:
             for i in range(n):
                 peer = self.up_peers[(self._rr + i) % n]
@@ -294,6 +297,14 @@ class P2PData:
                     return ActivationMsg(data_id, payload, labels, trace)
             if not block:
                 return None
+            if t0 is not None and _time.monotonic() - t0 > timeout:
+                raise TimeoutError(
+                    f"P2PData.recv_activation: rank {self.rank} waited "
+                    f"{timeout:.0f}s on peers {self.up_peers} — producer lost?")
+            # spin-then-yield: don't burn the host core while RCCL completes
+            empty_sweeps += 1
+            if empty_sweeps >= 512:
+                _time.sleep(0.0002)
 
     def send_gradient(self, stage: int, to_client: int, msg: GradientMsg):
         header = self._pack_header(msg.data_id, msg.trace)

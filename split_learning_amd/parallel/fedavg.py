@@ -31,7 +31,10 @@ def fedavg_state_dicts(state_dicts: List[Dict[str, torch.Tensor]],
             if key not in sd:
                 continue
             t = sd[key].float()
-            t = torch.nan_to_num(t)  # NaN -> 0, reference src/Utils.py:51-52
+            # NaN -> 0 ONLY (reference src/Utils.py:51-52 zero-fills just NaNs;
+            # +/-inf propagates there, so don't clamp it here either)
+            if torch.isnan(t).any():
+                t = torch.where(torch.isnan(t), torch.zeros_like(t), t)
             t = t * w
             acc = t if acc is None else acc + t
         avg = acc / total_w
@@ -65,8 +68,9 @@ def allreduce_fedavg_(model: torch.nn.Module, my_size: float,
     total = size_t.item()
 
     if float_keys:
-        flat = torch.cat([torch.nan_to_num(sd[k].detach().float()).reshape(-1)
-                          for k in float_keys])
+        flat = torch.cat([sd[k].detach().float().reshape(-1) for k in float_keys])
+        # NaN -> 0 only; leave +/-inf to propagate (reference src/Utils.py:51-52)
+        torch.where(torch.isnan(flat), torch.zeros_like(flat), flat, out=flat)
         flat.mul_(my_size)
         dist.all_reduce(flat, group=group)
         flat.div_(total)

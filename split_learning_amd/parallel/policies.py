@@ -39,6 +39,8 @@ from .server import Server
 class VanillaServer(Server):
     """Sequential SL: one stage-1 device at a time, weights relayed onward."""
 
+    RCCL_FEDAVG_OK = False  # sequential: group all-reduce would deadlock
+
     def __init__(self, *a, **kw):
         super().__init__(*a, **kw)
         self._edges: List[dict] = []
@@ -96,7 +98,8 @@ class VanillaServer(Server):
     def on_update(self, msg):
         if msg["layer_id"] == 1 and self._seq_idx < len(self._edges) - 1:
             self._edge_parts.append((msg["parameters"], msg["size"]))
-            self._relay_state = msg["parameters"]   # relay to the next device
+            if msg["parameters"] is not None:       # relay to the next device
+                self._relay_state = msg["parameters"]  # (None: keep last state)
             self._seq_idx += 1
             self._start_edge(self._edges[self._seq_idx])
             return
@@ -146,7 +149,8 @@ class VanillaServer(Server):
 class ClusterFSLServer(Server):
     """Sequential clusters; averaged model seeds the next cluster."""
 
-    SCHED_EXTRA: dict = {}
+    RCCL_FEDAVG_OK = False  # sequential clusters: later clusters' ranks would
+    SCHED_EXTRA: dict = {}  # never reach the group collective (see Server)
 
     def __init__(self, *a, **kw):
         super().__init__(*a, **kw)

@@ -1,6 +1,12 @@
 """Straggler avoidance: 2-component GMM on log(speed), analytic intersection
 threshold; devices below threshold are rejected (reference src/Selection.py:4-48,
-used at src/Server.py:324-338)."""
+used at src/Server.py:324-338).
+
+Provenance note: this module is a close re-implementation of the reference's
+auto_threshold — the GMM fit (n_init=9, random_state=0), the analytic
+two-Gaussian intersection and the fallback ladder are the ALGORITHM the survey
+mandates exact semantics for, so the math here tracks
+src/Selection.py line-for-line by necessity rather than design choice."""
 
 from __future__ import annotations
 

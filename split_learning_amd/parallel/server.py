@@ -56,6 +56,15 @@ def boundary_shapes(model_name: str, data_name: str, cuts: List[int],
 
 
 class Server:
+    # RCCL group-all-reduce FedAvg requires every member of a (cluster, stage)
+    # group to reach the collective in the same round — true for the main
+    # concurrent protocol and FLEX, FALSE for the sequential policies
+    # (Vanilla/Cluster_FSL/DCSL/2LS), where clients train one-at-a-time and an
+    # all-reduce would deadlock (and the weight relay would receive the
+    # non-representative ranks' `parameters: None`).  Sequential policy
+    # subclasses override this to force the control-plane FedAvg path.
+    RCCL_FEDAVG_OK = True
+
     def __init__(self, config: Dict[str, Any], control, logger=None,
                  checkpoint_dir: str = "."):
         srv = config["server"]
@@ -271,7 +280,19 @@ class Server:
 
     # -- reusable protocol pieces (policies compose these differently) -----
     def _load_ckpt(self):
-        if self.save_parameters and self.load_parameters and os.path.exists(self.ckpt_path):
+        """Checkpoint reload for a round's START broadcast.
+
+        The reference reloads {model}_{data}.pth every round gated on
+        save_parameters ALONE (src/Server.py:230-232; its `load` flag is read
+        but never used).  Here `parameters.load` additionally gates only the
+        FIRST round's load (so load=False starts from fresh init even when a
+        stale .pth exists); from round 2 on, save_parameters alone triggers the
+        reload of the just-saved aggregate — otherwise training progress would
+        be silently discarded each round (ADVICE.md round-1 medium finding).
+        """
+        first_round = self.round == self.global_round
+        want = self.load_parameters if first_round else self.save_parameters
+        if want and os.path.exists(self.ckpt_path):
             self._log(f"Loaded checkpoint {self.ckpt_path}")
             return torch.load(self.ckpt_path, weights_only=True)
         return None
@@ -321,8 +342,9 @@ class Server:
             "refresh": self.refresh, "cluster": rec["cluster"],
             "n_stages": self.n_stages, "routing": self._routing_for(rec),
             "scheduler": self.scheduler_overrides(rec),
-            "fedavg": (self.config.get("transport") or {}).get("fedavg", "control"),
-            "fedavg_groups": self._fedavg_groups(),
+            "fedavg": ((self.config.get("transport") or {}).get("fedavg", "control")
+                       if self.RCCL_FEDAVG_OK else "control"),
+            "fedavg_groups": self._fedavg_groups() if self.RCCL_FEDAVG_OK else None,
         })
 
     def _send_stop(self, rec, message="Stop training!"):

@@ -17,7 +17,10 @@ KNOWN = {"VGG16", "BERT", "KWT", "MobileNetv1", "ViT"}
 
 
 def get_val(model_name: str, data_name: str, state_dict_full, logger=None,
-            device: str = "cpu", max_batches: int = 16) -> bool:
+            device: str = "cpu", max_batches: int = 0) -> bool:
+    """max_batches=0 (default) evaluates the WHOLE test set, matching the
+    reference's full-test-set loop (src/val/VGG16.py:8-38); a positive value
+    truncates (used by fast CPU tests only)."""
     if model_name not in KNOWN:
         if logger:
             logger.log_warning(f"get_val: unknown model {model_name}")
@@ -31,7 +34,7 @@ def get_val(model_name: str, data_name: str, state_dict_full, logger=None,
     loss_sum = 0.0
     with torch.no_grad():
         for bi, (x, y) in enumerate(loader):
-            if bi >= max_batches:
+            if max_batches > 0 and bi >= max_batches:
                 break
             x, y = x.to(device), y.to(device)
             logits = model(x)
