@@ -257,17 +257,31 @@ class DistPipeline:
                                  depth=CONTROL_COUNT + 1)
         self.next_id = 1
 
+    # a wedged peer (bad comm init, lost message) surfaces as a named
+    # TimeoutError with the rank/peer in it instead of hanging the driver run
+    RECV_TIMEOUT_S = 300.0
+
     def run(self, steps):
         from split_learning_amd.ops import functional as hf
         from split_learning_amd.parallel.messages import ActivationMsg, GradientMsg
         import collections
+        phase_log = os.environ.get("SL_BENCH_PHASE_LOG", "0") == "1"
+        tp0 = time.perf_counter()
+
+        def plog(what):
+            if phase_log:
+                log(f"[bench rank {self.rank}] {what} at "
+                    f"+{time.perf_counter() - tp0:.3f}s")
+
         nan_flag = torch.zeros((), dtype=torch.bool, device=self.device)
+        plog(f"run({steps}) start")
         if self.is_first:
             xs, ys = make_batches(self.device, steps, seed=self.rank + self.next_id)
             inflight = collections.deque()
 
             def bwd_one():
-                g = self.plane.recv_gradient(1, self.rank, block=True)
+                g = self.plane.recv_gradient(1, self.rank, block=True,
+                                             timeout=self.RECV_TIMEOUT_S)
                 xo = inflight.popleft()
                 self.opt.zero_grad()
                 out = self.model(xo)
@@ -283,11 +297,14 @@ class DistPipeline:
                 inflight.append(xs[i])
                 if len(inflight) >= CONTROL_COUNT:
                     bwd_one()
+                if phase_log and (i + 1) % max(1, steps // 4) == 0:
+                    plog(f"fwd {i + 1}/{steps}")
             while inflight:
                 bwd_one()
         else:
             for i in range(steps):
-                m = self.plane.recv_activation(1, 0, self.rank, block=True)
+                m = self.plane.recv_activation(1, 0, self.rank, block=True,
+                                               timeout=self.RECV_TIMEOUT_S)
                 act = m.data.requires_grad_(True)
                 self.opt.zero_grad()
                 logits = self.model(act)
@@ -298,6 +315,9 @@ class DistPipeline:
                 self.opt.step()
                 self.plane.send_gradient(1, self.peer, GradientMsg(
                     m.data_id, act.grad.detach(), []))
+                if phase_log and (i + 1) % max(1, steps // 4) == 0:
+                    plog(f"step {i + 1}/{steps}")
+        plog(f"run({steps}) done")
         # the loss (and thus the NaN flag) lives on stage-2 ranks; surface it
         # on rank 0 where the JSON line is printed
         import torch.distributed as dist

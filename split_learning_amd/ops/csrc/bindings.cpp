@@ -8,10 +8,12 @@ at::Tensor matmul_f32(const at::Tensor&, const at::Tensor&, bool, bool,
 at::Tensor linear_fwd(const at::Tensor&, const at::Tensor&, c10::optional<at::Tensor>);
 at::Tensor colsum_f32(const at::Tensor&);
 // conv2d.hip
+at::Tensor pad_nchw(const at::Tensor&, int);
 at::Tensor conv2d_fwd(const at::Tensor&, const at::Tensor&, c10::optional<at::Tensor>,
-                      int, int);
+                      int, int, bool);
 at::Tensor conv2d_bwd_data(const at::Tensor&, const at::Tensor&, int, int, int, int);
-at::Tensor conv2d_bwd_weight(const at::Tensor&, const at::Tensor&, int, int, int, int);
+at::Tensor conv2d_bwd_weight(const at::Tensor&, const at::Tensor&, int, int, int, int,
+                             bool);
 at::Tensor conv2d_bwd_bias(const at::Tensor&);
 // norm.hip
 std::vector<at::Tensor> bn2d_stats_fused(const at::Tensor&,
@@ -74,9 +76,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out") = py::none(), py::arg("accumulate") = false);
   m.def("linear_fwd", &slk::linear_fwd);
   m.def("colsum_f32", &slk::colsum_f32);
-  m.def("conv2d_fwd", &slk::conv2d_fwd);
+  m.def("pad_nchw", &slk::pad_nchw);
+  m.def("conv2d_fwd", &slk::conv2d_fwd, py::arg("x"), py::arg("w"),
+        py::arg("bias"), py::arg("stride"), py::arg("pad"),
+        py::arg("x_is_padded") = false);
   m.def("conv2d_bwd_data", &slk::conv2d_bwd_data);
-  m.def("conv2d_bwd_weight", &slk::conv2d_bwd_weight);
+  m.def("conv2d_bwd_weight", &slk::conv2d_bwd_weight, py::arg("gy"), py::arg("x"),
+        py::arg("kh"), py::arg("kw"), py::arg("stride"), py::arg("pad"),
+        py::arg("x_is_padded") = false);
   m.def("conv2d_bwd_bias", &slk::conv2d_bwd_bias);
   m.def("bn2d_stats_fused", &slk::bn2d_stats_fused);
   m.def("bn2d_fwd", &slk::bn2d_fwd);

@@ -26,20 +26,22 @@ struct StridedGather {
   long sAb, sAm, sAk;
   long sBb, sBk, sBn;
 
+  struct KCtx {};
   struct ACtx { const float* p; bool valid; };
   struct BCtx { const float* p; bool valid; };
 
-  __device__ ACtx prepA(int b, int m, bool valid) const {
+  __device__ KCtx prepK(int) const { return {}; }
+  __device__ ACtx prepA(int b, int m, bool valid, int) const {
     return {A + (long)b * sAb + (long)m * sAm, valid};
   }
-  __device__ float loadA(const ACtx& c, int k, bool kv) const {
+  __device__ float loadA(const ACtx& c, const KCtx&, int k, bool kv) const {
     const float v = c.p[(long)k * sAk];
     return (c.valid & kv) ? v : 0.f;
   }
-  __device__ BCtx prepB(int b, int n, bool valid) const {
+  __device__ BCtx prepB(int b, int n, bool valid, int) const {
     return {B + (long)b * sBb + (long)n * sBn, valid};
   }
-  __device__ float loadB(const BCtx& c, int k, bool kv) const {
+  __device__ float loadB(const BCtx& c, const KCtx&, int k, bool kv) const {
     const float v = c.p[(long)k * sBk];
     return (c.valid & kv) ? v : 0.f;
   }

@@ -14,13 +14,20 @@
 // Gather contract (PMC-driven redesign — the first version's per-element
 // bounds branches compiled to 56 s_and_saveexec chains per loop and put waves
 // 53% issue-stalled, profiles/):
-//   * prepA(batch, m_clamped, valid) / prepB(batch, n_clamped, valid) run ONCE
-//     per thread (m/n are staging-loop invariants): they hoist the row/column
-//     address decomposition into a small ctx;
-//   * loadA(ctx, k)/loadB(ctx, k) must be BRANCHLESS: always load from a
-//     clamped in-bounds address and select 0 via the valid flags (cndmask,
-//     not exec-mask branches).  k arrives already clamped to [0, K-1];
-//     k_valid covers the K tail.
+//   * prepA(batch, m_clamped, valid, dk) / prepB(batch, n_clamped, valid, dk)
+//     run ONCE per thread (m/n are staging-loop invariants): they hoist the
+//     row/column address decomposition into a small ctx.  `dk` is the
+//     thread's CONSTANT intra-tile k offset (A staging is k-fast: every one
+//     of a thread's RA elements shares dk = tid & 15; B staging is n-fast:
+//     ctx i loads dk = (tid >> 6) + i * waves) — functors may fold dk into
+//     hoisted addresses;
+//   * prepK(k0) runs once per K-tile with a wave-uniform k0 and returns a
+//     small KCtx of SALU-computed per-tile scalars (e.g. the hoisted
+//     (batch, row) window decomposition of the round-2 conv gathers);
+//   * loadA(ctx, kctx, k)/loadB(ctx, kctx, k) must be BRANCHLESS: always load
+//     from a clamped in-bounds address and select 0 via the valid flags
+//     (cndmask, not exec-mask branches).  k arrives already clamped to
+//     [0, K-1]; k_valid covers the K tail.
 #pragma once
 
 #include "common.h"
@@ -94,30 +101,35 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
   constexpr int RB = (SLK_BN * SLK_BK) / 256;  // per-thread B elements
   float ra[RA], rb[RB];
 
-  // hoisted per-thread staging contexts (m/n fixed across the K loop)
+  // hoisted per-thread staging contexts (m/n fixed across the K loop).
+  // A staging is k-fast: (tid + i*256) & 15 == tid & 15 for every i, so ALL
+  // of a thread's A elements share one intra-tile k offset a_dk.
+  const int a_dk = tid & (SLK_BK - 1);
   typename Gather::ACtx actx[RA];
   typename Gather::BCtx bctx[RB];
-  int ka[RA], kb[RB];
+  int kb[RB];
   #pragma unroll
   for (int i = 0; i < RA; ++i) {
     const int idx = tid + i * 256;
     const int m = m0 + (idx >> 4);      // k-fast: contiguous global rows
-    ka[i] = idx & 15;
-    actx[i] = g.prepA(batch, min(m, M - 1), m < M);
+    actx[i] = g.prepA(batch, min(m, M - 1), m < M, a_dk);
   }
   #pragma unroll
   for (int i = 0; i < RB; ++i) {
     const int idx = tid + i * 256;
     const int n = n0 + (idx & 63);      // n-fast: coalesced for row-major B
     kb[i] = idx >> 6;
-    bctx[i] = g.prepB(batch, min(n, N - 1), n < N);
+    bctx[i] = g.prepB(batch, min(n, N - 1), n < N, kb[i]);
   }
 
   auto load_tile = [&](int k0) {
+    // prepK: wave-uniform per-tile scalars (SALU) shared by every load
+    const typename Gather::KCtx kc =
+        g.prepK(__builtin_amdgcn_readfirstlane(k0));
     #pragma unroll
     for (int i = 0; i < RA; ++i) {
-      const int k = k0 + ka[i];
-      ra[i] = g.loadA(actx[i], min(k, K - 1), k < k_end);
+      const int k = k0 + a_dk;
+      ra[i] = g.loadA(actx[i], kc, min(k, K - 1), k < k_end);
     }
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
@@ -127,7 +139,7 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
       // FastDiv multiplies) compiles to SALU ops instead of per-lane VALU —
       // the gather address math was 139 VALU per 16 MFMA in the loop body.
       const int k = __builtin_amdgcn_readfirstlane(k0 + kb[i]);
-      rb[i] = g.loadB(bctx[i], min(k, K - 1), k < k_end);
+      rb[i] = g.loadB(bctx[i], kc, min(k, K - 1), k < k_end);
     }
   };
 
@@ -255,35 +267,37 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
   constexpr int RB = (SLK_BN2 * SLK_BK) / 256;  // 8
   float ra[RA], rb[RB];
 
+  const int a_dk = tid & (SLK_BK - 1);
   typename Gather::ACtx actx[RA];
   typename Gather::BCtx bctx[RB];
-  int ka[RA], kb[RB];
+  int kb[RB];
   #pragma unroll
   for (int i = 0; i < RA; ++i) {
     const int idx = tid + i * 256;
     const int m = m0 + (idx >> 4);      // k-fast staging, 128 rows
-    ka[i] = idx & 15;
-    actx[i] = g.prepA(batch, min(m, M - 1), m < M);
+    actx[i] = g.prepA(batch, min(m, M - 1), m < M, a_dk);
   }
   #pragma unroll
   for (int i = 0; i < RB; ++i) {
     const int idx = tid + i * 256;
     const int n = n0 + (idx & 127);     // n-fast staging, 128 cols
     kb[i] = idx >> 7;
-    bctx[i] = g.prepB(batch, min(n, N - 1), n < N);
+    bctx[i] = g.prepB(batch, min(n, N - 1), n < N, kb[i]);
   }
 
   auto load_tile = [&](int k0) {
+    const typename Gather::KCtx kc =
+        g.prepK(__builtin_amdgcn_readfirstlane(k0));
     #pragma unroll
     for (int i = 0; i < RA; ++i) {
-      const int k = k0 + ka[i];
-      ra[i] = g.loadA(actx[i], min(k, K - 1), k < k_end);
+      const int k = k0 + a_dk;
+      ra[i] = g.loadA(actx[i], kc, min(k, K - 1), k < k_end);
     }
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
       // (tid + i*256) >> 7 is wave-uniform (see the 64x64 kernel's note)
       const int k = __builtin_amdgcn_readfirstlane(k0 + kb[i]);
-      rb[i] = g.loadB(bctx[i], min(k, K - 1), k < k_end);
+      rb[i] = g.loadB(bctx[i], kc, min(k, K - 1), k < k_end);
     }
   };
 

@@ -316,7 +316,24 @@ class P2PData:
                       timeout=None) -> Optional[GradientMsg]:
         if self._grad_ring is None:
             return None
-        got = self._grad_ring.wait() if block else self._grad_ring.poll()
+        if block and timeout:
+            import time as _time
+            t0 = _time.monotonic()
+            got = None
+            empty = 0
+            while got is None:
+                got = self._grad_ring.poll()
+                if got is None:
+                    if _time.monotonic() - t0 > timeout:
+                        raise TimeoutError(
+                            f"P2PData.recv_gradient: rank {self.rank} waited "
+                            f"{timeout:.0f}s on peer {self.down_peer} — "
+                            f"consumer lost?")
+                    empty += 1
+                    if empty >= 512:
+                        _time.sleep(0.0002)
+        else:
+            got = self._grad_ring.wait() if block else self._grad_ring.poll()
         if got is None:
             return None
         header, _, payload = got
