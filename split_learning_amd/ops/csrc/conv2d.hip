@@ -1,28 +1,28 @@
 // Implicit-GEMM NCHW fp32 convolution on the MFMA tile framework.
 //
-// Forward:     y[Co, (B,OH,OW)] = W[Co, (Ci,KH,KW)] @ im2col(x_pad)
-// Bwd-data:    gx = valid-conv(pad(gy), flipT(W))      (stride 1)
-//              gx[Ci, (B,H,W)] = W^T-gather @ col(gy)  (stride > 1 fallback)
-// Bwd-weight:  gw[Co, (Ci,KH,KW)] = gy-gather @ im2col(x_pad)^T
+// Forward:     y[Co, (B,OH,OW)] = W[Co, (Ci,KH,KW)] @ im2col(x)
+// Bwd-data:    gx[Ci, (B,H,W)]  = W^T-gather @ col(gy)   (stride-aware)
+// Bwd-weight:  gw[Co, (Ci,KH,KW)] = gy-gather @ im2col(x)^T
 //
-// Round-2 redesign (PMC round 1: conv GEMMs = 55% of the step, MFMA only
-// ~29% busy, waves issue-bound on the per-element gather math — VERDICT #1):
-// boundary arithmetic is ELIMINATED from the hot loops by padding inputs in
-// memory once per call (pad_nchw kernel) so every gather is a valid-conv
-// (pad 0) whose addresses decompose into
-//     [per-thread hoisted pointer] + [wave-uniform SALU offset]
-// with NO per-element bounds compare/clamp/select chain:
-// * fwd loadB:   px(n) + (ci*Hp + kh)*Wp + kw      — SALU offset, plain load;
-// * bwd-data:    gx = fwd-conv of pad(gy, KH-1-p) with flip-transposed
-//   weights (w^T flip kernel, ~us on <10 MB) -> same bounds-free gather;
-// * bwd-weight:  x_pad gather is all-SALU; the gy (A-side) per-lane k
-//   decomposition collapses via the prepK window hoist: when OH*OW % 16 == 0
-//   a BK=16 window stays inside one (b, rem0) row, so b/rem decompose ONCE
-//   per tile on the scalar unit (KCtx) and the per-lane address is
+// Round-2 findings (profiles/, conv_microbench + PMC on MI355X):
+// * the round-1 hypothesis "gathers are issue-bound" was WRONG at kernel
+//   grain: a bounds-free pre-padded gather (SLK_CONV_PAD=1 path below) ties
+//   the bounds-checked direct gather within noise (42.2 vs 42.7 us on
+//   conv1_2) because the core is barrier/latency-paced, not VALU-bound —
+//   while its pad/flipT prep kernels cost ~0.27 ms/step.  The DIRECT gather
+//   is therefore the default; the padded path stays selectable for A/B as
+//   the core gets faster.
+// * the real levers are in tile_gemm.h (LDS ping-pong, batched reads) and
+//   split-K slab stores (AtomicStore at split 64 = 4M atomicAdds on a 65 KB
+//   output: the 512ch/2x2 tail layers ran 2x slower than MIOpen).
+// * bwd-weight keeps one genuine round-2 win: the gy (A-side) per-lane k
+//   decomposition collapses via the prepK window hoist — when OH*OW % 16
+//   == 0 a BK=16 window stays inside one (b, rem0) row, so b/rem decompose
+//   once per tile on the scalar unit and the per-lane address is
 //   row_m + dk + kctx.a_off.
-// Geometry (KH, KW, stride) stays a TEMPLATE specialisation; runtime dims
-// divide through FastDiv magics (common.h); every conv GEMM split-Ks when its
-// tile grid underfills the chip (fp32 atomic accumulation, first-split bias).
+// Geometry (KH, KW, stride, pad) is a TEMPLATE specialisation for the model
+// zoo's cases; runtime dims divide through FastDiv magics (common.h);
+// gathers are BRANCHLESS (clamped addresses + cndmask selects).
 #include <torch/extension.h>
 #include <ATen/ATen.h>
 #include <c10/hip/HIPStream.h>
@@ -40,19 +40,27 @@ static inline at::Tensor zeroed(at::IntArrayRef sizes, const at::TensorOptions& 
   return t;
 }
 
+static inline bool conv_pad_mode() {
+  static int v = [] {
+    const char* e = std::getenv("SLK_CONV_PAD");
+    return e ? atoi(e) : 0;
+  }();
+  return v != 0;
+}
+
 
 struct ConvGeom {
   int B, Ci, H, W, Co, KH, KW, OH, OW, stride, pad;
   FastDiv d_ohow, d_ow, d_hw, d_w, d_khkw, d_kw;  // runtime-dim magics
 };
 
-// compile-time geometry: CKH=0 means runtime (generic fallback)
-template <int CKH, int CKW, int CS, int CP>
+// compile-time geometry: CKH=0 means runtime (generic fallback); pad is
+// always runtime (one add in hoisted prep code — not worth a template axis)
+template <int CKH, int CKW, int CS>
 struct Geo {
   static constexpr bool fixed = CKH > 0;
   __device__ static int kh_kw(const ConvGeom& g) { return fixed ? CKH * CKW : g.KH * g.KW; }
   __device__ static int stride(const ConvGeom& g) { return fixed ? CS : g.stride; }
-  __device__ static int pad(const ConvGeom& g) { return fixed ? CP : g.pad; }
   // k -> (c, kh, kw) with strength-reduced division when fixed
   __device__ static void dk(const ConvGeom& g, int k, int& c, int& kh, int& kw) {
     if (fixed) {
@@ -71,13 +79,55 @@ struct Geo {
 
 __device__ __forceinline__ float sel0(float v, bool keep) { return keep ? v : 0.f; }
 
-// ---------------- forward (pre-padded input: pad == 0 in-kernel) ----------
+// ---------------- forward: direct gather (default) ------------------------
 template <int CKH, int CKW, int CS>
 struct ConvFwdGather {
-  using G = Geo<CKH, CKW, CS, 0>;
-  const float* w;  // [Co, Ci, KH, KW] (or [Ci, Co, KHKW] flipT for bwd-data)
-  const float* x;  // [B, Ci, H, W] pre-padded: every tap address is in-bounds
+  using G = Geo<CKH, CKW, CS>;
+  const float* w;  // [Co, Ci, KH, KW]
+  const float* x;  // [B, Ci, H, W]
   ConvGeom geo;
+
+  struct KCtx {};
+  __device__ KCtx prepK(int) const { return {}; }
+
+  struct ACtx { const float* row; bool valid; };
+  struct BCtx { const float* base; int ihb, iwb; bool valid; };
+
+  __device__ ACtx prepA(int, int m, bool valid, int) const {
+    return {w + (long)m * (geo.Ci * G::kh_kw(geo)), valid};
+  }
+  __device__ float loadA(const ACtx& c, const KCtx&, int k, bool kv) const {
+    return sel0(c.row[k], c.valid & kv);
+  }
+  __device__ BCtx prepB(int, int n, bool valid, int) const {
+    const unsigned b = geo.d_ohow.div(n);
+    const unsigned rem = geo.d_ohow.mod(n, b);
+    const unsigned oh = geo.d_ow.div(rem);
+    const unsigned ow = geo.d_ow.mod(rem, oh);
+    return {x + (long)b * geo.Ci * geo.H * geo.W,
+            (int)oh * G::stride(geo) - geo.pad,
+            (int)ow * G::stride(geo) - geo.pad, valid};
+  }
+  __device__ float loadB(const BCtx& c, const KCtx&, int k, bool kv) const {
+    int ci, kh, kw;
+    G::dk(geo, k, ci, kh, kw);
+    const int ih = c.ihb + kh;
+    const int iw = c.iwb + kw;
+    const bool in = (unsigned)ih < (unsigned)geo.H && (unsigned)iw < (unsigned)geo.W;
+    const int ihc = in ? ih : 0;
+    const int iwc = in ? iw : 0;
+    const float v = c.base[((long)ci * geo.H + ihc) * geo.W + iwc];
+    return sel0(v, c.valid & kv & in);
+  }
+};
+
+// ---------------- forward: pre-padded bounds-free gather (SLK_CONV_PAD) ---
+template <int CKH, int CKW, int CS>
+struct ConvFwdGatherP {
+  using G = Geo<CKH, CKW, CS>;
+  const float* w;  // [Co, Ci, KH, KW] (or [Ci, Co, KHKW] flipT for bwd-data)
+  const float* x;  // [B, Ci, Hp, Wp] pre-padded: every tap address in-bounds
+  ConvGeom geo;    // H/W are the PADDED dims; geo.pad == 0
 
   struct KCtx {};
   __device__ KCtx prepK(int) const { return {}; }
@@ -100,26 +150,25 @@ struct ConvFwdGather {
               + (long)(oh * G::stride(geo)) * geo.W + ow * G::stride(geo)};
   }
   __device__ float loadB(const BCtx& c, const KCtx&, int k, bool kv) const {
-    // k is wave-uniform: the whole decomposition + offset is SALU; the load
-    // is hoisted-pointer + scalar offset with NO bounds math (valid-conv)
     int ci, kh, kw;
-    G::dk(geo, k, ci, kh, kw);
+    G::dk(geo, k, ci, kh, kw);   // SALU (k wave-uniform)
     const int off = (ci * geo.H + kh) * geo.W + kw;
     return sel0(c.px[off], kv);
   }
 };
 
 struct ConvFwdStore {
-  float* y;  // [B, Co, OH, OW]
+  float* y;  // [B, Co, OH, OW] (or a [split_k, ...] slab when slab_stride > 0)
   const float* bias;  // [Co] nullable
   int Co, OHOW;
   FastDiv d_ohow;
   bool accumulate;
-  __device__ void store(int, int m, int n, float v, bool first_split) const {
+  long slab_stride;   // 0: direct store; else per-split slab offset (numel)
+  __device__ void store(int, int m, int n, float v, int ks) const {
     const unsigned b = d_ohow.div(n);
     const unsigned rem = d_ohow.mod(n, b);
-    if (bias != nullptr && first_split) v += bias[m];
-    float* p = y + ((long)b * Co + m) * OHOW + rem;
+    if (bias != nullptr && ks == 0) v += bias[m];
+    float* p = y + (long)ks * slab_stride + ((long)b * Co + m) * OHOW + rem;
     if (accumulate) {
       atomicAdd(p, v);
     } else {
@@ -128,13 +177,10 @@ struct ConvFwdStore {
   }
 };
 
-// ---------------- backward data (stride > 1 fallback only) ----------------
-// stride-1 bwd-data runs as a forward valid-conv on pad(gy) with flipT(w);
-// this gather keeps the stride-aware scatter logic for the s=2/s=4 cases
-// (MobileNet 3x3 s2, ViT 4x4 s4), operating on the UNPADDED gy.
-template <int CKH, int CKW, int CS, int CP>
+// ---------------- backward data (direct stride-aware gather) ---------------
+template <int CKH, int CKW, int CS>
 struct ConvBwdDataGather {
-  using G = Geo<CKH, CKW, CS, CP>;
+  using G = Geo<CKH, CKW, CS>;
   const float* w;   // [Co, Ci, KH, KW]
   const float* gy;  // [B, Co, OH, OW]
   ConvGeom geo;
@@ -159,7 +205,7 @@ struct ConvBwdDataGather {
     const unsigned ih = geo.d_w.div(rem);
     const unsigned iw = geo.d_w.mod(rem, ih);
     return {gy + (long)b * geo.Co * geo.OH * geo.OW,
-            (int)ih + G::pad(geo), (int)iw + G::pad(geo), valid};
+            (int)ih + geo.pad, (int)iw + geo.pad, valid};
   }
   __device__ float loadB(const BCtx& c, const KCtx&, int k, bool kv) const {
     int co, kh, kw;
@@ -169,7 +215,11 @@ struct ConvBwdDataGather {
     const int ow_num = c.iwp - kw;
     int oh, ow;
     bool ok;
-    if (s == 2) {
+    if (s == 1) {
+      oh = oh_num;
+      ow = ow_num;
+      ok = true;
+    } else if (s == 2) {
       ok = ((oh_num | ow_num) & 1) == 0;
       oh = oh_num >> 1;
       ow = ow_num >> 1;
@@ -191,14 +241,15 @@ struct ConvBwdDataGather {
 };
 
 struct ConvBwdDataStore {
-  float* gx;  // [B, Ci, H, W]
+  float* gx;  // [B, Ci, H, W] (or slab)
   int Ci, HW;
   FastDiv d_hw;
   bool accumulate;
-  __device__ void store(int, int m, int n, float v, bool) const {
+  long slab_stride;
+  __device__ void store(int, int m, int n, float v, int ks) const {
     const unsigned b = d_hw.div(n);
     const unsigned rem = d_hw.mod(n, b);
-    float* p = gx + ((long)b * Ci + m) * HW + rem;
+    float* p = gx + (long)ks * slab_stride + ((long)b * Ci + m) * HW + rem;
     if (accumulate) {
       atomicAdd(p, v);
     } else {
@@ -207,17 +258,17 @@ struct ConvBwdDataStore {
   }
 };
 
-// ---------------- backward weight (pre-padded input) -----------------------
-template <int CKH, int CKW, int CS>
+// ---------------- backward weight ------------------------------------------
+// Direct gather; the gy A-side uses the prepK window hoist (fast flag).
+// PADDED=true additionally drops the x bounds math (SLK_CONV_PAD path).
+template <int CKH, int CKW, int CS, bool PADDED>
 struct ConvBwdWeightGather {
-  using G = Geo<CKH, CKW, CS, 0>;
+  using G = Geo<CKH, CKW, CS>;
   const float* gy;  // [B, Co, OH, OW]
-  const float* x;   // [B, Ci, H, W] pre-padded
+  const float* x;   // [B, Ci, H, W] (padded dims when PADDED)
   ConvGeom geo;
-  bool fast;        // OH*OW % 16 == 0: a BK window stays inside one (b, row)
+  bool fast;        // OH*OW % 16 == 0: window-hoisted A addressing
 
-  // window hoist: k = (b, oh, ow) decomposes ONCE per tile on the SALU when
-  // fast (k0 % 16 == 0 and OHOW % 16 == 0 keep b constant over the window)
   struct KCtx { long a_off; };
   __device__ KCtx prepK(int k0) const {
     if (!fast) return {0};
@@ -242,31 +293,84 @@ struct ConvBwdWeightGather {
     return sel0(v, c.valid & kv);
   }
 
-  struct BCtx { long off; };     // (ci*Hp + kh)*Wp + kw, hoisted per thread
-  __device__ BCtx prepB(int, int n, bool, int) const {
+  struct BCtx { long ciHW; int kh, kw; bool valid; };
+  __device__ BCtx prepB(int, int n, bool valid, int) const {
     int ci, kh, kw;
     G::dk(geo, n, ci, kh, kw);
-    return {((long)ci * geo.H + kh) * geo.W + kw};
+    return {(long)ci * geo.H * geo.W, kh, kw, valid};
   }
   __device__ float loadB(const BCtx& c, const KCtx&, int k, bool kv) const {
-    // all-SALU scalar offset (k wave-uniform), bounds-free on x_pad
     const unsigned b = geo.d_ohow.div(k);
     const unsigned rem = geo.d_ohow.mod(k, b);
     const unsigned oh = geo.d_ow.div(rem);
     const unsigned ow = geo.d_ow.mod(rem, oh);
-    const long sc = (long)b * geo.Ci * geo.H * geo.W
-                  + (long)(oh * G::stride(geo)) * geo.W + ow * G::stride(geo);
-    return sel0(x[c.off + sc], kv);
+    if (PADDED) {
+      const long sc = (long)b * geo.Ci * geo.H * geo.W
+                    + (long)(oh * G::stride(geo)) * geo.W + ow * G::stride(geo);
+      return sel0(x[c.ciHW + c.kh * geo.W + c.kw + sc], kv);
+    }
+    const int ih = (int)oh * G::stride(geo) - geo.pad + c.kh;
+    const int iw = (int)ow * G::stride(geo) - geo.pad + c.kw;
+    const bool in = (unsigned)ih < (unsigned)geo.H && (unsigned)iw < (unsigned)geo.W;
+    const int ihc = in ? ih : 0;
+    const int iwc = in ? iw : 0;
+    const float v = x[(long)b * geo.Ci * geo.H * geo.W + c.ciHW
+                      + (long)ihc * geo.W + iwc];
+    return sel0(v, c.valid & kv & in);
   }
 };
 
 struct AtomicStore {
-  float* c;  // [M, N] contiguous, pre-zeroed
+  float* c;  // [M, N] contiguous (pre-zeroed when accumulate) or slab
   int N;
-  __device__ void store(int, int m, int n, float v, bool) const {
-    atomicAdd(c + (long)m * N + n, v);
+  bool accumulate;
+  long slab_stride;
+  __device__ void store(int, int m, int n, float v, int ks) const {
+    float* p = c + (long)ks * slab_stride + (long)m * N + n;
+    if (accumulate) {
+      atomicAdd(p, v);
+    } else {
+      *p = v;
+    }
   }
 };
+
+// ---------------- split-K slab finalize ------------------------------------
+// out[i] = sum_s slab[s*numel + i]: replaces (zero + atomic accumulate) with
+// streaming slab writes + one reduction pass — at split 64 on a 65 KB output
+// the atomic path was 4M serialized RMWs.
+__global__ void slab_reduce_kernel(const float* __restrict__ slab, int splits,
+                                   float* __restrict__ out, long numel) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < numel;
+       i += stride) {
+    float s = 0.f;
+    for (int k = 0; k < splits; ++k) s += slab[(long)k * numel + i];
+    out[i] = s;
+  }
+}
+
+static void slab_reduce(const at::Tensor& slab, int splits, at::Tensor& out,
+                        hipStream_t stream) {
+  const long numel = out.numel();
+  const int grid = (int)std::min<long>((numel + 255) / 256, 4096);
+  hipLaunchKernelGGL(slab_reduce_kernel, dim3(grid), dim3(256), 0, stream,
+                     slab.data_ptr<float>(), splits, out.data_ptr<float>(),
+                     numel);
+}
+
+// slab mode threshold: slab stores beat atomics once several splits pile on
+// one output (env-tunable for A/B sweeps)
+static inline int slab_min_splits() {
+  static int v = [] {
+    const char* e = std::getenv("SLK_SLAB_MIN");
+    return e ? atoi(e) : 4;
+  }();
+  return v < 2 ? 2 : v;
+}
+
+// cap slab memory (splits * numel * 4B); above this stay with atomics
+static constexpr long SLK_SLAB_MAX_BYTES = 512l << 20;
 
 // ---------------- padding / weight-flip prep kernels -----------------------
 
@@ -338,24 +442,6 @@ static at::Tensor flipT_w(const at::Tensor& w) {
 
 // ---------------- host wrappers ----------------
 
-// geometry over the PADDED input (pad == 0 in-kernel)
-static ConvGeom make_geom_padded(int B, int Ci, int Hp, int Wp, int Co, int KH,
-                                 int KW, int stride) {
-  ConvGeom g;
-  g.B = B; g.Ci = Ci; g.H = Hp; g.W = Wp; g.Co = Co; g.KH = KH; g.KW = KW;
-  g.stride = stride; g.pad = 0;
-  g.OH = (Hp - KH) / stride + 1;
-  g.OW = (Wp - KW) / stride + 1;
-  g.d_ohow.init(g.OH * g.OW);
-  g.d_ow.init(g.OW);
-  g.d_hw.init(g.H * g.W);
-  g.d_w.init(g.W);
-  g.d_khkw.init(g.KH * g.KW);
-  g.d_kw.init(g.KW);
-  return g;
-}
-
-// geometry over the ORIGINAL input (bwd-data stride>1 fallback only)
 static ConvGeom make_geom(int B, int Ci, int H, int W, int Co, int KH, int KW,
                           int stride, int pad) {
   ConvGeom g;
@@ -372,7 +458,8 @@ static ConvGeom make_geom(int B, int Ci, int H, int W, int Co, int KH, int KW,
   return g;
 }
 
-// dispatch over the model zoo's conv geometries (padded: stride only)
+// dispatch over the model zoo's conv geometries (stride/tap shape only; pad
+// is runtime)
 template <typename F>
 static void dispatch_geom(const ConvGeom& g, F&& f) {
   if (g.KH == 3 && g.KW == 3 && g.stride == 1) {
@@ -401,33 +488,82 @@ struct PickGather<Gather, 3> { using type = Gather<4, 4, 4>; };
 template <template <int, int, int> class Gather>
 struct PickGather<Gather, 4> { using type = Gather<0, 0, 0>; };
 
-// x_is_padded: caller already ran pad_nchw (functional.py saves x_pad from
-// the forward so the backward-weight pass reuses it without re-padding)
+// bwd-weight needs the extra PADDED axis
+template <int A, int B, int C> using BwdWD = ConvBwdWeightGather<A, B, C, false>;
+template <int A, int B, int C> using BwdWP = ConvBwdWeightGather<A, B, C, true>;
+
+// split-K output plan: direct store (split 1), atomic accumulate (few
+// splits), or per-split slab + reduce (many splits)
+struct SplitPlan {
+  int split_k;
+  bool slab;
+  at::Tensor buf;       // the tensor the GEMM writes (slab or the output)
+  at::Tensor out;       // the real output
+};
+
+static SplitPlan plan_split(int split_k, at::IntArrayRef out_sizes,
+                            const at::TensorOptions& opt) {
+  SplitPlan p;
+  p.split_k = split_k;
+  long numel = 1;
+  for (auto s : out_sizes) numel *= s;
+  p.slab = split_k >= slab_min_splits() &&
+           (long)split_k * numel * 4 <= SLK_SLAB_MAX_BYTES;
+  if (p.slab) {
+    p.out = at::empty(out_sizes, opt);
+    std::vector<long> ss{p.split_k};
+    for (auto s : out_sizes) ss.push_back(s);
+    p.buf = at::empty(ss, opt);
+  } else if (split_k > 1) {
+    p.out = zeroed(out_sizes, opt);
+    p.buf = p.out;
+  } else {
+    p.out = at::empty(out_sizes, opt);
+    p.buf = p.out;
+  }
+  return p;
+}
+
+// x_is_padded: caller already ran pad_nchw (only meaningful under
+// SLK_CONV_PAD=1, where functional.py saves x_pad from the forward)
 at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                       c10::optional<at::Tensor> bias, int stride, int pad,
                       bool x_is_padded) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.scalar_type() == at::kFloat);
   TORCH_CHECK(w.is_cuda() && w.dim() == 4 && w.size(1) == x.size(1));
-  auto xp = x_is_padded ? x.contiguous() : pad_nchw(x, pad);
+  const bool padded = x_is_padded || conv_pad_mode();
+  auto stream = c10::hip::getCurrentHIPStream().stream();
   auto wc = w.contiguous();
-  ConvGeom geo = make_geom_padded(xp.size(0), xp.size(1), xp.size(2), xp.size(3),
-                                  w.size(0), w.size(2), w.size(3), stride);
+
+  at::Tensor xp = padded ? (x_is_padded ? x.contiguous() : pad_nchw(x, pad))
+                         : x.contiguous();
+  ConvGeom geo = padded
+      ? make_geom(xp.size(0), xp.size(1), xp.size(2), xp.size(3), w.size(0),
+                  w.size(2), w.size(3), stride, 0)
+      : make_geom(x.size(0), x.size(1), x.size(2), x.size(3), w.size(0),
+                  w.size(2), w.size(3), stride, pad);
   const int M = geo.Co, N = geo.B * geo.OH * geo.OW, K = geo.Ci * geo.KH * geo.KW;
   const int split_k = slk_pick_split_k(M, N, K, 1);
-  auto y = split_k > 1
-      ? zeroed({geo.B, geo.Co, geo.OH, geo.OW}, x.options())
-      : at::empty({geo.B, geo.Co, geo.OH, geo.OW}, x.options());
+  auto plan = plan_split(split_k, {geo.B, geo.Co, geo.OH, geo.OW}, x.options());
 
-  ConvFwdStore st{y.data_ptr<float>(),
+  ConvFwdStore st{plan.buf.data_ptr<float>(),
                   bias.has_value() ? bias->data_ptr<float>() : nullptr,
-                  geo.Co, geo.OH * geo.OW, geo.d_ohow, split_k > 1};
-  auto stream = c10::hip::getCurrentHIPStream().stream();
+                  geo.Co, geo.OH * geo.OW, geo.d_ohow,
+                  !plan.slab && split_k > 1,
+                  plan.slab ? plan.out.numel() : 0};
   dispatch_geom(geo, [&](auto ic) {
-    using GT = typename PickGather<ConvFwdGather, decltype(ic)::value>::type;
-    GT g{wc.data_ptr<float>(), xp.data_ptr<float>(), geo};
-    slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
+    if (padded) {
+      using GT = typename PickGather<ConvFwdGatherP, decltype(ic)::value>::type;
+      GT g{wc.data_ptr<float>(), xp.data_ptr<float>(), geo};
+      slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
+    } else {
+      using GT = typename PickGather<ConvFwdGather, decltype(ic)::value>::type;
+      GT g{wc.data_ptr<float>(), xp.data_ptr<float>(), geo};
+      slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
+    }
   });
-  return y;
+  if (plan.slab) slab_reduce(plan.buf, split_k, plan.out, stream);
+  return plan.out;
 }
 
 at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride,
@@ -438,30 +574,29 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
   const int Ci = w.size(1), Co = w.size(0);
   const int B = gy.size(0);
 
-  if (stride == 1 && KH == KW && pad <= KH - 1) {
+  if (conv_pad_mode() && stride == 1 && KH == KW && pad <= KH - 1) {
     // gx = valid-conv(pad(gy, K-1-p), flipT(w)): bounds-free forward gather.
-    // Derivation: y[oh] sums x[oh - p + kh]  =>  gx[ih] = sum_kh gy[ih + p - kh]
+    // Derivation: y[oh] sums x[oh - p + kh] => gx[ih] = sum_kh gy[ih + p - kh]
     // = sum_kh' gy_pad[ih + kh'] with kh' = KH-1-kh and pad q = KH-1-p.
     auto wt = flipT_w(w);
     const int q = KH - 1 - pad;
     auto gyp = pad_nchw(gy, q);
-    ConvGeom geo = make_geom_padded(B, Co, gyp.size(2), gyp.size(3), Ci, KH, KW, 1);
+    ConvGeom geo = make_geom(B, Co, gyp.size(2), gyp.size(3), Ci, KH, KW, 1, 0);
     TORCH_CHECK(geo.OH == H && geo.OW == W, "bwd-data geometry mismatch");
     const int M = Ci, N = B * H * W, K = Co * KH * KW;
     const int split_k = slk_pick_split_k(M, N, K, 1);
-    auto gx = split_k > 1 ? zeroed({B, Ci, H, W}, gy.options())
-                          : at::empty({B, Ci, H, W}, gy.options());
-    ConvFwdStore st{gx.data_ptr<float>(), nullptr, Ci, H * W, geo.d_ohow,
-                    split_k > 1};
+    auto plan = plan_split(split_k, {B, Ci, H, W}, gy.options());
+    ConvFwdStore st{plan.buf.data_ptr<float>(), nullptr, Ci, H * W, geo.d_ohow,
+                    !plan.slab && split_k > 1, plan.slab ? plan.out.numel() : 0};
     dispatch_geom(geo, [&](auto ic) {
-      using GT = typename PickGather<ConvFwdGather, decltype(ic)::value>::type;
+      using GT = typename PickGather<ConvFwdGatherP, decltype(ic)::value>::type;
       GT g{wt.data_ptr<float>(), gyp.data_ptr<float>(), geo};
       slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
     });
-    return gx;
+    if (plan.slab) slab_reduce(plan.buf, split_k, plan.out, stream);
+    return plan.out;
   }
 
-  // stride > 1 (or non-square taps): scatter-style gather on the unpadded gy
   auto gyc = gy.contiguous();
   auto wc = w.contiguous();
   ConvGeom geo = make_geom(B, Ci, H, W, Co, KH, KW, stride, pad);
@@ -469,45 +604,56 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
               "bwd-data geometry mismatch");
   const int M = geo.Ci, N = geo.B * H * W, K = geo.Co * KH * KW;
   const int split_k = slk_pick_split_k(M, N, K, 1);
-  auto gx = split_k > 1 ? zeroed({B, Ci, H, W}, gy.options())
-                        : at::empty({B, Ci, H, W}, gy.options());
-  ConvBwdDataStore st{gx.data_ptr<float>(), geo.Ci, H * W, geo.d_hw,
-                      split_k > 1};
-  if (KH == 3 && KW == 3 && stride == 2 && pad == 1) {
-    ConvBwdDataGather<3, 3, 2, 1> g{wc.data_ptr<float>(), gyc.data_ptr<float>(), geo};
+  auto plan = plan_split(split_k, {B, Ci, H, W}, gy.options());
+  ConvBwdDataStore st{plan.buf.data_ptr<float>(), geo.Ci, H * W, geo.d_hw,
+                      !plan.slab && split_k > 1,
+                      plan.slab ? plan.out.numel() : 0};
+  dispatch_geom(geo, [&](auto ic) {
+    using GT = typename PickGather<ConvBwdDataGather, decltype(ic)::value>::type;
+    GT g{wc.data_ptr<float>(), gyc.data_ptr<float>(), geo};
     slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
-  } else if (KH == 4 && KW == 4 && stride == 4 && pad == 0) {
-    ConvBwdDataGather<4, 4, 4, 0> g{wc.data_ptr<float>(), gyc.data_ptr<float>(), geo};
-    slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
-  } else {
-    ConvBwdDataGather<0, 0, 0, 0> g{wc.data_ptr<float>(), gyc.data_ptr<float>(), geo};
-    slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
-  }
-  return gx;
+  });
+  if (plan.slab) slab_reduce(plan.buf, split_k, plan.out, stream);
+  return plan.out;
 }
 
 at::Tensor conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x, int KH,
                              int KW, int stride, int pad, bool x_is_padded) {
   TORCH_CHECK(gy.is_cuda() && x.is_cuda() && gy.scalar_type() == at::kFloat);
+  const bool padded = x_is_padded || conv_pad_mode();
   auto gyc = gy.contiguous();
-  auto xp = x_is_padded ? x.contiguous() : pad_nchw(x, pad);
-  ConvGeom geo = make_geom_padded(xp.size(0), xp.size(1), xp.size(2), xp.size(3),
-                                  gy.size(1), KH, KW, stride);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+
+  at::Tensor xp = padded ? (x_is_padded ? x.contiguous() : pad_nchw(x, pad))
+                         : x.contiguous();
+  ConvGeom geo = padded
+      ? make_geom(xp.size(0), xp.size(1), xp.size(2), xp.size(3), gy.size(1),
+                  KH, KW, stride, 0)
+      : make_geom(x.size(0), x.size(1), x.size(2), x.size(3), gy.size(1),
+                  KH, KW, stride, pad);
   TORCH_CHECK(geo.OH == gy.size(2) && geo.OW == gy.size(3),
               "bwd-weight geometry mismatch");
   const int M = geo.Co, N = geo.Ci * KH * KW, K = geo.B * geo.OH * geo.OW;
-  auto gw = zeroed({geo.Co, geo.Ci, KH, KW}, gy.options());
-
-  AtomicStore st{gw.data_ptr<float>(), N};
   int split_k = slk_pick_split_k(M, N, K, 1);
+  auto plan = plan_split(split_k, {geo.Co, geo.Ci, KH, KW}, gy.options());
+  // bwd-weight always accumulates over K even at split 1 unless slab? No:
+  // at split 1 a single block owns each output tile -> direct store is fine.
+  AtomicStore st{plan.buf.data_ptr<float>(), N, !plan.slab && split_k > 1,
+                 plan.slab ? plan.out.numel() : 0};
   const bool fast = (geo.OH * geo.OW) % SLK_BK == 0;
-  auto stream = c10::hip::getCurrentHIPStream().stream();
   dispatch_geom(geo, [&](auto ic) {
-    using GT = typename PickGather<ConvBwdWeightGather, decltype(ic)::value>::type;
-    GT g{gyc.data_ptr<float>(), xp.data_ptr<float>(), geo, fast};
-    slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
+    if (padded) {
+      using GT = typename PickGather<BwdWP, decltype(ic)::value>::type;
+      GT g{gyc.data_ptr<float>(), xp.data_ptr<float>(), geo, fast};
+      slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
+    } else {
+      using GT = typename PickGather<BwdWD, decltype(ic)::value>::type;
+      GT g{gyc.data_ptr<float>(), xp.data_ptr<float>(), geo, fast};
+      slk_launch_gemm(g, st, M, N, K, 1, split_k, stream);
+    }
   });
-  return gw;
+  if (plan.slab) slab_reduce(plan.buf, split_k, plan.out, stream);
+  return plan.out;
 }
 
 // per-channel sum of gy over (B, OH, OW) -> conv bias gradient

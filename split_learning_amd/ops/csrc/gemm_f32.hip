@@ -53,9 +53,9 @@ struct StridedStore {
   int N;
   const float* bias;  // per-column bias (nullable)
   bool accumulate;    // atomic accumulate (split-K or beta=1)
-  __device__ void store(int b, int m, int n, float v, bool first_split) const {
+  __device__ void store(int b, int m, int n, float v, int ks) const {
     float* p = C + (long)b * sCb + (long)m * N + n;
-    if (bias != nullptr && first_split) v += bias[n];
+    if (bias != nullptr && ks == 0) v += bias[n];
     if (accumulate) {
       atomicAdd(p, v);
     } else {

@@ -64,8 +64,13 @@ __device__ __forceinline__ int slk_lds_phys(int k, int m) {
 template <typename Gather, typename Store>
 __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     Gather g, Store st, int M, int N, int K, int split_k, int k_per_split) {
-  __shared__ __align__(16) float ldsA[SLK_BK * SLK_LDS_ROW];
-  __shared__ __align__(16) float ldsB[SLK_BK * SLK_LDS_ROW];
+  // round-2 core restructure (PMC: 53% issue-stall + 32% barrier-parked,
+  // MFMA pipe only ~37% busy): LDS is PING-PONG double-buffered so the loop
+  // runs ONE barrier per BK tile (stage into buf p^1 while reading buf p),
+  // and each tile's 8 ds_read_b64 issue as one up-front batch so the 16
+  // MFMAs go back-to-back behind a single lgkmcnt wait.
+  __shared__ __align__(16) float ldsA[2][SLK_BK * SLK_LDS_ROW];
+  __shared__ __align__(16) float ldsB[2][SLK_BK * SLK_LDS_ROW];
 
   const int tile_n = blockIdx.x;
   const int tile_m = blockIdx.y;
@@ -143,20 +148,26 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     }
   };
 
-  load_tile(k_begin);
-  for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
+  auto stage_to = [&](int p) {
     #pragma unroll
     for (int i = 0; i < RA; ++i) {
       const int idx = tid + i * 256;
-      ldsA[slk_lds_phys(idx & 15, idx >> 4)] = ra[i];
+      ldsA[p][slk_lds_phys(idx & 15, idx >> 4)] = ra[i];
     }
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
       const int idx = tid + i * 256;
-      ldsB[slk_lds_phys(idx >> 6, idx & 63)] = rb[i];
+      ldsB[p][slk_lds_phys(idx >> 6, idx & 63)] = rb[i];
     }
-    __syncthreads();
-    if (k0 + SLK_BK < k_end) load_tile(k0 + SLK_BK);  // overlaps the MFMAs
+  };
+
+  load_tile(k_begin);
+  stage_to(0);
+  __syncthreads();
+  int p = 0;
+  for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
+    const bool has_next = k0 + SLK_BK < k_end;
+    if (has_next) load_tile(k0 + SLK_BK);  // global loads overlap the MFMAs
 
 #ifdef SLK_MFMA32
     // operand map (ISA): lane l supplies A[i=l&31][k=l>>5], B[k=l>>5][j=l&31]
@@ -164,27 +175,36 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     const int l31 = lane & 31;
     #pragma unroll
     for (int kk = 0; kk < SLK_BK / 2; ++kk) {
-      float a = ldsA[slk_lds_phys(kk * 2 + kh, wm + l31)];
-      float b = ldsB[slk_lds_phys(kk * 2 + kh, wn + l31)];
+      float a = ldsA[p][slk_lds_phys(kk * 2 + kh, wm + l31)];
+      float b = ldsB[p][slk_lds_phys(kk * 2 + kh, wn + l31)];
       acc32 = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc32, 0, 0, 0);
     }
 #else
     // operand pairs (frag_c, frag_c+16) are adjacent dwords in the swizzled
-    // layout -> each f32x2 load is one conflict-free ds_read_b64
+    // layout -> each f32x2 load is one conflict-free ds_read_b64; all 8 reads
+    // first, then the 16 MFMAs back-to-back
+    f32x2 a01[SLK_BK / 4], b01[SLK_BK / 4];
     #pragma unroll
     for (int kk = 0; kk < SLK_BK / 4; ++kk) {
       const int kr = kk * 4 + frag_r;
-      f32x2 a01 = *reinterpret_cast<const f32x2*>(
-          &ldsA[slk_lds_phys(kr, wm + frag_c)]);
-      f32x2 b01 = *reinterpret_cast<const f32x2*>(
-          &ldsB[slk_lds_phys(kr, wn + frag_c)]);
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.x, b01.x, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.x, b01.y, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.y, b01.x, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.y, b01.y, acc[1][1], 0, 0, 0);
+      a01[kk] = *reinterpret_cast<const f32x2*>(
+          &ldsA[p][slk_lds_phys(kr, wm + frag_c)]);
+      b01[kk] = *reinterpret_cast<const f32x2*>(
+          &ldsB[p][slk_lds_phys(kr, wn + frag_c)]);
+    }
+    #pragma unroll
+    for (int kk = 0; kk < SLK_BK / 4; ++kk) {
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01[kk].x, b01[kk].x, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01[kk].x, b01[kk].y, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01[kk].y, b01[kk].x, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01[kk].y, b01[kk].y, acc[1][1], 0, 0, 0);
     }
 #endif
+    if (has_next) stage_to(p ^ 1);  // write the OTHER buffer: no read hazard
+    // one barrier per tile: publishes buf p^1's writes AND closes buf p's
+    // reads before it is overwritten next iteration
     __syncthreads();
+    p ^= 1;
   }
 
 #ifdef SLK_MFMA32
@@ -194,7 +214,7 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
   for (int r = 0; r < 16; ++r) {
     int row = m0 + wm + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
     int col = n0 + wn + (lane & 31);
-    if (row < M && col < N) st.store(batch, row, col, acc32[r], ks == 0);
+    if (row < M && col < N) st.store(batch, row, col, acc32[r], ks);
   }
 #else
   // epilogue: C/D fragment mapping for 16x16x4: col = lane&15, row = (lane>>4)*4 + i
@@ -206,7 +226,7 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
       for (int i = 0; i < 4; ++i) {
         int row = m0 + wm + mi * 16 + frag_r * 4 + i;
         int col = n0 + wn + ni * 16 + frag_c;
-        if (row < M && col < N) st.store(batch, row, col, acc[mi][ni][i], ks == 0);
+        if (row < M && col < N) st.store(batch, row, col, acc[mi][ni][i], ks);
       }
     }
   }
@@ -341,7 +361,7 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
       for (int r = 0; r < 16; ++r) {
         int row = m0 + wm + mi * 32 + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
         int col = n0 + wn + ni * 32 + (lane & 31);
-        if (row < M && col < N) st.store(batch, row, col, acc[mi][ni][r], ks == 0);
+        if (row < M && col < N) st.store(batch, row, col, acc[mi][ni][r], ks);
       }
     }
   }

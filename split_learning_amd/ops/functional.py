@@ -135,37 +135,32 @@ def linear(x, weight, bias=None):
 # ---------------------------------------------------------------------------
 
 class Conv2dFn(Function):
-    """Padded-input implicit-GEMM conv: the forward pads x once (pad_nchw)
-    and SAVES the padded tensor so both the forward gather and the
-    backward-weight gather run bounds-free (conv2d.hip round-2 redesign);
-    backward-data runs as a forward valid-conv on pad(gy) with flip-
-    transposed weights (stride 1) or the stride-aware fallback."""
+    """Implicit-GEMM conv on the MFMA tile framework.  The default path
+    gathers directly from the unpadded tensors (the bounds math measured
+    free — conv2d.hip round-2 notes); SLK_CONV_PAD=1 switches the C++ side
+    to the pre-padded bounds-free gathers for A/B experiments."""
 
     @staticmethod
     def forward(ctx, x, weight, bias, stride, padding):
         x = x.contiguous()
-        xp = native().pad_nchw(x, padding) if padding > 0 else x
-        y = native().conv2d_fwd(xp, weight, bias, stride, padding,
-                                x_is_padded=True)
-        ctx.save_for_backward(xp, weight)
+        y = native().conv2d_fwd(x, weight, bias, stride, padding)
+        ctx.save_for_backward(x, weight)
         ctx.stride = stride
         ctx.padding = padding
-        ctx.in_hw = (x.shape[2], x.shape[3])
         ctx.has_bias = bias is not None
         return y
 
     @staticmethod
     def backward(ctx, gy):
-        xp, weight = ctx.saved_tensors
+        x, weight = ctx.saved_tensors
         gy = gy.contiguous()
         gx = gw = gb = None
         if ctx.needs_input_grad[0]:
             gx = native().conv2d_bwd_data(gy, weight, ctx.stride, ctx.padding,
-                                          ctx.in_hw[0], ctx.in_hw[1])
+                                          x.shape[2], x.shape[3])
         if ctx.needs_input_grad[1]:
-            gw = native().conv2d_bwd_weight(gy, xp, weight.shape[2], weight.shape[3],
-                                            ctx.stride, ctx.padding,
-                                            x_is_padded=True)
+            gw = native().conv2d_bwd_weight(gy, x, weight.shape[2], weight.shape[3],
+                                            ctx.stride, ctx.padding)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             gb = native().conv2d_bwd_bias(gy)
         return gx, gw, gb, None, None
