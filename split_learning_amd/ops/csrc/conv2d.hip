@@ -543,7 +543,7 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
       : make_geom(x.size(0), x.size(1), x.size(2), x.size(3), w.size(0),
                   w.size(2), w.size(3), stride, pad);
   const int M = geo.Co, N = geo.B * geo.OH * geo.OW, K = geo.Ci * geo.KH * geo.KW;
-  const int split_k = slk_pick_split_k(M, N, K, 1);
+  const int split_k = slk_effective_split(K, slk_pick_split_k(M, N, K, 1));
   auto plan = plan_split(split_k, {geo.B, geo.Co, geo.OH, geo.OW}, x.options());
 
   ConvFwdStore st{plan.buf.data_ptr<float>(),
@@ -584,7 +584,7 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
     ConvGeom geo = make_geom(B, Co, gyp.size(2), gyp.size(3), Ci, KH, KW, 1, 0);
     TORCH_CHECK(geo.OH == H && geo.OW == W, "bwd-data geometry mismatch");
     const int M = Ci, N = B * H * W, K = Co * KH * KW;
-    const int split_k = slk_pick_split_k(M, N, K, 1);
+    const int split_k = slk_effective_split(K, slk_pick_split_k(M, N, K, 1));
     auto plan = plan_split(split_k, {B, Ci, H, W}, gy.options());
     ConvFwdStore st{plan.buf.data_ptr<float>(), nullptr, Ci, H * W, geo.d_ohow,
                     !plan.slab && split_k > 1, plan.slab ? plan.out.numel() : 0};
@@ -603,7 +603,7 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
   TORCH_CHECK(geo.OH == gy.size(2) && geo.OW == gy.size(3),
               "bwd-data geometry mismatch");
   const int M = geo.Ci, N = geo.B * H * W, K = geo.Co * KH * KW;
-  const int split_k = slk_pick_split_k(M, N, K, 1);
+  const int split_k = slk_effective_split(K, slk_pick_split_k(M, N, K, 1));
   auto plan = plan_split(split_k, {B, Ci, H, W}, gy.options());
   ConvBwdDataStore st{plan.buf.data_ptr<float>(), geo.Ci, H * W, geo.d_hw,
                       !plan.slab && split_k > 1,
@@ -634,7 +634,7 @@ at::Tensor conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x, int KH,
   TORCH_CHECK(geo.OH == gy.size(2) && geo.OW == gy.size(3),
               "bwd-weight geometry mismatch");
   const int M = geo.Co, N = geo.Ci * KH * KW, K = geo.B * geo.OH * geo.OW;
-  int split_k = slk_pick_split_k(M, N, K, 1);
+  const int split_k = slk_effective_split(K, slk_pick_split_k(M, N, K, 1));
   auto plan = plan_split(split_k, {geo.Co, geo.Ci, KH, KW}, gy.options());
   // bwd-weight always accumulates over K even at split 1 unless slab? No:
   // at split 1 a single block owns each output tile -> direct store is fine.

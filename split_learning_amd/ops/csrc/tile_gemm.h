@@ -61,16 +61,19 @@ __device__ __forceinline__ int slk_lds_phys(int k, int m) {
          (((2 * (m & 15)) | ((m >> 4) & 1)) ^ (2 * k));
 }
 
-template <typename Gather, typename Store>
+// PP (ping-pong) axis of the round-2 core experiment (PMC: 53% issue-stall +
+// 32% barrier-parked, MFMA pipe only ~37% busy):
+//   PP=true : LDS double-buffered, ONE barrier per BK tile (stage into buf
+//             p^1 while reading buf p) — but 24.6 KB LDS caps the CU at 6
+//             blocks (6 waves/SIMD);
+//   PP=false: single-buffer two-barrier loop, 12.3 KB LDS, register-limited
+//             7 waves/SIMD.
+// Both compile; SLK_PP picks at launch so one GPU session can A/B them.
+template <typename Gather, typename Store, bool PP>
 __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     Gather g, Store st, int M, int N, int K, int split_k, int k_per_split) {
-  // round-2 core restructure (PMC: 53% issue-stall + 32% barrier-parked,
-  // MFMA pipe only ~37% busy): LDS is PING-PONG double-buffered so the loop
-  // runs ONE barrier per BK tile (stage into buf p^1 while reading buf p),
-  // and each tile's 8 ds_read_b64 issue as one up-front batch so the 16
-  // MFMAs go back-to-back behind a single lgkmcnt wait.
-  __shared__ __align__(16) float ldsA[2][SLK_BK * SLK_LDS_ROW];
-  __shared__ __align__(16) float ldsB[2][SLK_BK * SLK_LDS_ROW];
+  __shared__ __align__(16) float ldsA[PP ? 2 : 1][SLK_BK * SLK_LDS_ROW];
+  __shared__ __align__(16) float ldsB[PP ? 2 : 1][SLK_BK * SLK_LDS_ROW];
 
   const int tile_n = blockIdx.x;
   const int tile_m = blockIdx.y;
@@ -108,24 +111,20 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
 
   // hoisted per-thread staging contexts (m/n fixed across the K loop).
   // A staging is k-fast: (tid + i*256) & 15 == tid & 15 for every i, so ALL
-  // of a thread's A elements share one intra-tile k offset a_dk.
+  // of a thread's A elements share one intra-tile k offset a_dk.  B staging
+  // is n-fast with (tid + i*256) & 63 == tid & 63: every B element of a
+  // thread shares ONE column context (a single BCtx, not RB copies — the
+  // duplicate contexts cost ~10 VGPR and a wave/SIMD of occupancy).
   const int a_dk = tid & (SLK_BK - 1);
   typename Gather::ACtx actx[RA];
-  typename Gather::BCtx bctx[RB];
-  int kb[RB];
   #pragma unroll
   for (int i = 0; i < RA; ++i) {
     const int idx = tid + i * 256;
     const int m = m0 + (idx >> 4);      // k-fast: contiguous global rows
     actx[i] = g.prepA(batch, min(m, M - 1), m < M, a_dk);
   }
-  #pragma unroll
-  for (int i = 0; i < RB; ++i) {
-    const int idx = tid + i * 256;
-    const int n = n0 + (idx & 63);      // n-fast: coalesced for row-major B
-    kb[i] = idx >> 6;
-    bctx[i] = g.prepB(batch, min(n, N - 1), n < N, kb[i]);
-  }
+  const int nb = n0 + (tid & 63);       // n-fast: coalesced for row-major B
+  typename Gather::BCtx bctx = g.prepB(batch, min(nb, N - 1), nb < N, tid >> 6);
 
   auto load_tile = [&](int k0) {
     // prepK: wave-uniform per-tile scalars (SALU) shared by every load
@@ -138,13 +137,13 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     }
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
-      // kb[i] = (tid + i*256) >> 6 is WAVE-UNIFORM by construction (the >>6
+      // (tid + i*256) >> 6 is WAVE-UNIFORM by construction (the >>6
       // collapses the 64 lanes of a wave to its wave id); readfirstlane makes
       // that provable so the functor's whole k-decomposition (divisions,
       // FastDiv multiplies) compiles to SALU ops instead of per-lane VALU —
       // the gather address math was 139 VALU per 16 MFMA in the loop body.
-      const int k = __builtin_amdgcn_readfirstlane(k0 + kb[i]);
-      rb[i] = g.loadB(bctx[i], kc, min(k, K - 1), k < k_end);
+      const int k = __builtin_amdgcn_readfirstlane(k0 + (tid >> 6) + i * 4);
+      rb[i] = g.loadB(bctx, kc, min(k, K - 1), k < k_end);
     }
   };
 
@@ -161,14 +160,7 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     }
   };
 
-  load_tile(k_begin);
-  stage_to(0);
-  __syncthreads();
-  int p = 0;
-  for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
-    const bool has_next = k0 + SLK_BK < k_end;
-    if (has_next) load_tile(k0 + SLK_BK);  // global loads overlap the MFMAs
-
+  auto do_mfma = [&](int p) {
 #ifdef SLK_MFMA32
     // operand map (ISA): lane l supplies A[i=l&31][k=l>>5], B[k=l>>5][j=l&31]
     const int kh = lane >> 5;
@@ -181,30 +173,47 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
     }
 #else
     // operand pairs (frag_c, frag_c+16) are adjacent dwords in the swizzled
-    // layout -> each f32x2 load is one conflict-free ds_read_b64; all 8 reads
-    // first, then the 16 MFMAs back-to-back
-    f32x2 a01[SLK_BK / 4], b01[SLK_BK / 4];
+    // layout -> each f32x2 load is one conflict-free ds_read_b64.
+    // (Batching all 8 reads up front measured SLOWER: +16 VGPR pushed the
+    // allocation over the 64-register step, 8 -> 6 waves/SIMD.)
     #pragma unroll
     for (int kk = 0; kk < SLK_BK / 4; ++kk) {
       const int kr = kk * 4 + frag_r;
-      a01[kk] = *reinterpret_cast<const f32x2*>(
+      f32x2 a01 = *reinterpret_cast<const f32x2*>(
           &ldsA[p][slk_lds_phys(kr, wm + frag_c)]);
-      b01[kk] = *reinterpret_cast<const f32x2*>(
+      f32x2 b01 = *reinterpret_cast<const f32x2*>(
           &ldsB[p][slk_lds_phys(kr, wn + frag_c)]);
-    }
-    #pragma unroll
-    for (int kk = 0; kk < SLK_BK / 4; ++kk) {
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01[kk].x, b01[kk].x, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01[kk].x, b01[kk].y, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01[kk].y, b01[kk].x, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01[kk].y, b01[kk].y, acc[1][1], 0, 0, 0);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.x, b01.x, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.x, b01.y, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.y, b01.x, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a01.y, b01.y, acc[1][1], 0, 0, 0);
     }
 #endif
-    if (has_next) stage_to(p ^ 1);  // write the OTHER buffer: no read hazard
-    // one barrier per tile: publishes buf p^1's writes AND closes buf p's
-    // reads before it is overwritten next iteration
+  };
+
+  load_tile(k_begin);
+  if (PP) {
+    stage_to(0);
     __syncthreads();
-    p ^= 1;
+    int p = 0;
+    for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
+      const bool has_next = k0 + SLK_BK < k_end;
+      if (has_next) load_tile(k0 + SLK_BK);  // global loads overlap the MFMAs
+      do_mfma(p);
+      if (has_next) stage_to(p ^ 1);  // write the OTHER buffer: no hazard
+      // one barrier per tile: publishes buf p^1's writes AND closes buf p's
+      // reads before it is overwritten next iteration
+      __syncthreads();
+      p ^= 1;
+    }
+  } else {
+    for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
+      stage_to(0);
+      __syncthreads();
+      if (k0 + SLK_BK < k_end) load_tile(k0 + SLK_BK);
+      do_mfma(0);
+      __syncthreads();
+    }
   }
 
 #ifdef SLK_MFMA32
@@ -289,21 +298,14 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
 
   const int a_dk = tid & (SLK_BK - 1);
   typename Gather::ACtx actx[RA];
-  typename Gather::BCtx bctx[RB];
-  int kb[RB];
   #pragma unroll
   for (int i = 0; i < RA; ++i) {
     const int idx = tid + i * 256;
     const int m = m0 + (idx >> 4);      // k-fast staging, 128 rows
     actx[i] = g.prepA(batch, min(m, M - 1), m < M, a_dk);
   }
-  #pragma unroll
-  for (int i = 0; i < RB; ++i) {
-    const int idx = tid + i * 256;
-    const int n = n0 + (idx & 127);     // n-fast staging, 128 cols
-    kb[i] = idx >> 7;
-    bctx[i] = g.prepB(batch, min(n, N - 1), n < N, kb[i]);
-  }
+  const int nb = n0 + (tid & 127);      // n-fast staging, 128 cols
+  typename Gather::BCtx bctx = g.prepB(batch, min(nb, N - 1), nb < N, tid >> 7);
 
   auto load_tile = [&](int k0) {
     const typename Gather::KCtx kc =
@@ -316,8 +318,8 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
       // (tid + i*256) >> 7 is wave-uniform (see the 64x64 kernel's note)
-      const int k = __builtin_amdgcn_readfirstlane(k0 + kb[i]);
-      rb[i] = g.loadB(bctx[i], kc, min(k, K - 1), k < k_end);
+      const int k = __builtin_amdgcn_readfirstlane(k0 + (tid >> 7) + i * 2);
+      rb[i] = g.loadB(bctx, kc, min(k, K - 1), k < k_end);
     }
   };
 
@@ -376,6 +378,25 @@ inline bool slk_use_big(int M, int N) {
   return v != 0 && M >= 96 && N >= 96;
 }
 
+// BK-alignment can REDUCE the effective split count (K=4608 at split 64 ->
+// k_per_split 80 -> 58 splits).  Callers sizing per-split slabs must use this
+// exact value or trailing slab slices stay uninitialized.
+inline int slk_effective_split(int K, int split_k) {
+  if (split_k < 1) return 1;
+  int k_per_split = ceil_div(K, split_k);
+  k_per_split = ((k_per_split + SLK_BK - 1) / SLK_BK) * SLK_BK;
+  return ceil_div(K, k_per_split);
+}
+
+// SLK_PP=0 selects the single-buffer two-barrier loop (A/B axis; default 1)
+inline bool slk_pp_mode() {
+  static int v = [] {
+    const char* e = std::getenv("SLK_PP");
+    return e ? atoi(e) : 1;
+  }();
+  return v != 0;
+}
+
 // Host-side launch helper.
 template <typename Gather, typename Store>
 inline void slk_launch_gemm(const Gather& g, const Store& st, int M, int N, int K,
@@ -393,8 +414,13 @@ inline void slk_launch_gemm(const Gather& g, const Store& st, int M, int N, int 
     return;
   }
   dim3 grid(ceil_div(N, SLK_BN), ceil_div(M, SLK_BM), n_batch * split_k);
-  hipLaunchKernelGGL((slk_mfma_gemm_kernel<Gather, Store>), grid, dim3(256), 0,
-                     stream, g, st, M, N, K, split_k, k_per_split);
+  if (slk_pp_mode()) {
+    hipLaunchKernelGGL((slk_mfma_gemm_kernel<Gather, Store, true>), grid,
+                       dim3(256), 0, stream, g, st, M, N, K, split_k, k_per_split);
+  } else {
+    hipLaunchKernelGGL((slk_mfma_gemm_kernel<Gather, Store, false>), grid,
+                       dim3(256), 0, stream, g, st, M, N, K, split_k, k_per_split);
+  }
 }
 
 // Heuristic: pick split_k so the grid oversubscribes the 256 CUs (~5 blocks/CU:
