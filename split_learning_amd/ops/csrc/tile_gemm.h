@@ -71,11 +71,20 @@ __device__ __forceinline__ int slk_lds_phys(int k, int m) {
 // Both compile; SLK_PP picks at launch so one GPU session can A/B them.
 template <typename Gather, typename Store, bool PP>
 __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel(
-    Gather g, Store st, int M, int N, int K, int split_k, int k_per_split) {
+    Gather g, Store st, int M, int N, int K, int split_k, int k_per_split,
+    int xcd_remap) {
   __shared__ __align__(16) float ldsA[PP ? 2 : 1][SLK_BK * SLK_LDS_ROW];
   __shared__ __align__(16) float ldsB[PP ? 2 : 1][SLK_BK * SLK_LDS_ROW];
 
-  const int tile_n = blockIdx.x;
+  // XCD-aware n-tile placement: consecutive blockIdx.x land on XCD b % 8
+  // with disjoint per-XCD L2s, so the linear map sprays one image's B-side
+  // gathers across all 8 L2s.  When the n-grid is a multiple of 8, remap so
+  // XCD k owns the CONTIGUOUS n-range [k*T/8, (k+1)*T/8): bijective, and the
+  // per-XCD working set of x drops 8x (one L2 can then hold its slice).
+  int tile_n = blockIdx.x;
+  if (xcd_remap && (gridDim.x & 7) == 0) {
+    tile_n = (blockIdx.x & 7) * (gridDim.x >> 3) + (blockIdx.x >> 3);
+  }
   const int tile_m = blockIdx.y;
   const int batch = blockIdx.z / split_k;
   const int ks = blockIdx.z % split_k;
@@ -388,11 +397,14 @@ inline int slk_effective_split(int K, int split_k) {
   return ceil_div(K, k_per_split);
 }
 
-// SLK_PP=0 selects the single-buffer two-barrier loop (A/B axis; default 1)
+// SLK_PP=1 selects the ping-pong single-barrier loop.  Measured on MI355X
+// (profiles/SUMMARY.md round 2): PP loses everywhere — its 24.6 KB LDS caps
+// the CU at 6 blocks vs the two-barrier loop's register-limited 7 waves/SIMD,
+// and occupancy beats barrier elision (bench 10297 vs 9448 img/s).
 inline bool slk_pp_mode() {
   static int v = [] {
     const char* e = std::getenv("SLK_PP");
-    return e ? atoi(e) : 1;
+    return e ? atoi(e) : 0;
   }();
   return v != 0;
 }
@@ -414,12 +426,18 @@ inline void slk_launch_gemm(const Gather& g, const Store& st, int M, int N, int 
     return;
   }
   dim3 grid(ceil_div(N, SLK_BN), ceil_div(M, SLK_BM), n_batch * split_k);
+  static const int xcd = [] {
+    const char* e = std::getenv("SLK_XCD");
+    return e ? atoi(e) : 1;
+  }();
   if (slk_pp_mode()) {
     hipLaunchKernelGGL((slk_mfma_gemm_kernel<Gather, Store, true>), grid,
-                       dim3(256), 0, stream, g, st, M, N, K, split_k, k_per_split);
+                       dim3(256), 0, stream, g, st, M, N, K, split_k,
+                       k_per_split, xcd);
   } else {
     hipLaunchKernelGGL((slk_mfma_gemm_kernel<Gather, Store, false>), grid,
-                       dim3(256), 0, stream, g, st, M, N, K, split_k, k_per_split);
+                       dim3(256), 0, stream, g, st, M, N, K, split_k,
+                       k_per_split, xcd);
   }
 }
 
