@@ -63,17 +63,25 @@ class SequentialUnits(PartitionedModel):
     def forward(self, x):
         # stage-executor fusion: an adjacent BatchNorm2d -> ReLU pair inside
         # one partition runs as a single fused kernel on GPU (the cut can
-        # still split the pair, in which case both run standalone)
-        from ..ops.modules import HipBatchNorm2d, HipReLU
+        # still split the pair, in which case both run standalone); a Conv2d
+        # feeding a training-mode BatchNorm skips its bias-gradient reduction
+        # (analytically zero through batch normalisation — ops/functional.py)
+        from ..ops.modules import HipBatchNorm2d, HipConv2d, HipReLU
         units = self.active_units()
         n = len(units)
         i = 0
         while i < n:
             mod = getattr(self, f"layer{units[i]}")
-            if (isinstance(mod, HipBatchNorm2d) and x.is_cuda and i + 1 < n
-                    and isinstance(getattr(self, f"layer{units[i + 1]}"), HipReLU)):
+            nxt = getattr(self, f"layer{units[i + 1]}") if i + 1 < n else None
+            if (isinstance(mod, HipBatchNorm2d) and x.is_cuda
+                    and isinstance(nxt, HipReLU)):
                 x = mod(x, fuse_relu=True)
                 i += 2
+                continue
+            if (isinstance(mod, HipConv2d) and x.is_cuda and self.training
+                    and isinstance(nxt, HipBatchNorm2d) and nxt.training):
+                x = mod(x, bias_grad_zero=True)
+                i += 1
                 continue
             x = mod(x)
             i += 1

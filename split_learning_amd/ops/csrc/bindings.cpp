@@ -18,15 +18,16 @@ at::Tensor conv2d_bwd_bias(const at::Tensor&);
 // norm.hip
 std::vector<at::Tensor> bn2d_stats_fused(const at::Tensor&,
                                          c10::optional<at::Tensor>,
+                                         c10::optional<at::Tensor>,
                                          c10::optional<at::Tensor>, double, double);
 at::Tensor bn2d_fwd(const at::Tensor&, const at::Tensor&, const at::Tensor&,
                     const at::Tensor&, const at::Tensor&, bool);
 std::vector<at::Tensor> bn2d_bwd(const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&,
-                                 const at::Tensor&);
+                                 const at::Tensor&, c10::optional<at::Tensor>);
 std::vector<at::Tensor> bn2d_bwd_eval(const at::Tensor&, const at::Tensor&,
                                       const at::Tensor&, const at::Tensor&,
-                                      const at::Tensor&);
+                                      const at::Tensor&, c10::optional<at::Tensor>);
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor&, const at::Tensor&,
                                       const at::Tensor&, double);
 std::vector<at::Tensor> layernorm_bwd(const at::Tensor&, const at::Tensor&,
@@ -87,8 +88,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_bias", &slk::conv2d_bwd_bias);
   m.def("bn2d_stats_fused", &slk::bn2d_stats_fused);
   m.def("bn2d_fwd", &slk::bn2d_fwd);
-  m.def("bn2d_bwd", &slk::bn2d_bwd);
-  m.def("bn2d_bwd_eval", &slk::bn2d_bwd_eval);
+  m.def("bn2d_bwd", &slk::bn2d_bwd, py::arg("x"), py::arg("gy"),
+        py::arg("gamma"), py::arg("mean"), py::arg("invstd"),
+        py::arg("relu_y") = py::none());
+  m.def("bn2d_bwd_eval", &slk::bn2d_bwd_eval, py::arg("x"), py::arg("gy"),
+        py::arg("gamma"), py::arg("mean"), py::arg("invstd"),
+        py::arg("relu_y") = py::none());
   m.def("layernorm_fwd", &slk::layernorm_fwd);
   m.def("layernorm_bwd", &slk::layernorm_bwd);
   m.def("relu_fwd", &slk::relu_fwd);

@@ -60,9 +60,12 @@ __global__ void bn_finalize_kernel(const float* __restrict__ slab, int chunks,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
-                                   float* __restrict__ running_var, int C, float n,
+                                   float* __restrict__ running_var,
+                                   long* __restrict__ nbt, int C, float n,
                                    float momentum, float eps) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  // num_batches_tracked rides along (was a separate aten long-add launch)
+  if (c == 0 && nbt != nullptr) nbt[0] += 1;
   if (c >= C) return;
   float s = 0.f, s2 = 0.f;
   for (int k = 0; k < chunks; ++k) {
@@ -98,8 +101,12 @@ __global__ void bn_fwd_kernel(const float* __restrict__ x, float* __restrict__ y
 // reductions for backward: sum(gy) and sum(gy * xhat) per channel, chunked
 // over grid.y like the forward stats (one block per channel leaves most CUs
 // idle at C=64)
+// relu_y != nullptr fuses the preceding ReLU's backward mask (g = y>0 ? gy
+// : 0) into the reduction — the standalone relu_bwd launch and its full
+// read+write pass disappear from the BN+ReLU block backward.
 __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
                                      const float* __restrict__ gy,
+                                     const float* __restrict__ relu_y,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
                                      float* __restrict__ slab, int B, int C,
@@ -116,7 +123,9 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
     const int b = i / HW;
     const int r = i - b * HW;
     const long off = ((long)b * C + c) * HW + r;
-    const double g = (double)gy[off];
+    float gf = gy[off];
+    if (relu_y != nullptr && relu_y[off] <= 0.f) gf = 0.f;
+    const double g = (double)gf;
     s += g;
     sx += g * (double)((x[off] - m) * is);
   }
@@ -146,6 +155,7 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ slab, int chunk
 
 __global__ void bn_bwd_dx_kernel(const float* __restrict__ x,
                                  const float* __restrict__ gy,
+                                 const float* __restrict__ relu_y,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ invstd,
                                  const float* __restrict__ gamma,
@@ -157,12 +167,14 @@ __global__ void bn_bwd_dx_kernel(const float* __restrict__ x,
        i += (long)gridDim.x * blockDim.x) {
     const int c = (int)((i / HW) % C);
     const float is = invstd[c];
+    float g = gy[i];
+    if (relu_y != nullptr && relu_y[i] <= 0.f) g = 0.f;
     if (training) {
       const float xhat = (x[i] - mean[c]) * is;
       gx[i] = gamma[c] * is *
-              (gy[i] - sum_gy[c] * inv_n - xhat * sum_gy_xhat[c] * inv_n);
+              (g - sum_gy[c] * inv_n - xhat * sum_gy_xhat[c] * inv_n);
     } else {
-      gx[i] = gy[i] * gamma[c] * is;
+      gx[i] = g * gamma[c] * is;
     }
   }
 }
@@ -177,6 +189,7 @@ static inline int bn_chunks(long per_channel) {
 std::vector<at::Tensor> bn2d_stats_fused(const at::Tensor& x,
                                          c10::optional<at::Tensor> running_mean,
                                          c10::optional<at::Tensor> running_var,
+                                         c10::optional<at::Tensor> num_batches_tracked,
                                          double momentum, double eps) {
   const int B = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   auto mean = at::empty({C}, x.options());
@@ -193,6 +206,8 @@ std::vector<at::Tensor> bn2d_stats_fused(const at::Tensor& x,
                                               : nullptr,
                      running_var.has_value() ? running_var->data_ptr<float>()
                                              : nullptr,
+                     num_batches_tracked.has_value()
+                         ? num_batches_tracked->data_ptr<long>() : nullptr,
                      C, (float)((long)B * HW), (float)momentum, (float)eps);
   return {mean, invstd};
 }
@@ -216,7 +231,9 @@ at::Tensor bn2d_fwd(const at::Tensor& x, const at::Tensor& mean,
 static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tensor& gy,
                                              const at::Tensor& gamma,
                                              const at::Tensor& mean,
-                                             const at::Tensor& invstd, bool training) {
+                                             const at::Tensor& invstd, bool training,
+                                             const c10::optional<at::Tensor>& relu_y) {
+  const float* ry = relu_y.has_value() ? relu_y->data_ptr<float>() : nullptr;
   const int B = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   const long total = x.numel();
   auto sum_gy = at::empty({C}, x.options());
@@ -227,7 +244,7 @@ static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tens
   auto slab = at::empty({2, rchunks, C}, x.options());
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C, rchunks),
                      dim3(256), 0, stream,
-                     x.data_ptr<float>(), gy.data_ptr<float>(),
+                     x.data_ptr<float>(), gy.data_ptr<float>(), ry,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      slab.data_ptr<float>(), B, C, HW);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
@@ -235,7 +252,7 @@ static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tens
                      sum_gy.data_ptr<float>(), sum_gy_xhat.data_ptr<float>(), C);
   int grid = (int)std::min<long>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(grid), dim3(256), 0, stream,
-                     x.data_ptr<float>(), gy.data_ptr<float>(),
+                     x.data_ptr<float>(), gy.data_ptr<float>(), ry,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      gamma.data_ptr<float>(), sum_gy.data_ptr<float>(),
                      sum_gy_xhat.data_ptr<float>(), gx.data_ptr<float>(), total, C,
@@ -246,14 +263,16 @@ static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tens
 
 std::vector<at::Tensor> bn2d_bwd(const at::Tensor& x, const at::Tensor& gy,
                                  const at::Tensor& gamma, const at::Tensor& mean,
-                                 const at::Tensor& invstd) {
-  return bn2d_bwd_impl(x, gy, gamma, mean, invstd, true);
+                                 const at::Tensor& invstd,
+                                 c10::optional<at::Tensor> relu_y) {
+  return bn2d_bwd_impl(x, gy, gamma, mean, invstd, true, relu_y);
 }
 
 std::vector<at::Tensor> bn2d_bwd_eval(const at::Tensor& x, const at::Tensor& gy,
                                       const at::Tensor& gamma, const at::Tensor& mean,
-                                      const at::Tensor& invstd) {
-  return bn2d_bwd_impl(x, gy, gamma, mean, invstd, false);
+                                      const at::Tensor& invstd,
+                                      c10::optional<at::Tensor> relu_y) {
+  return bn2d_bwd_impl(x, gy, gamma, mean, invstd, false, relu_y);
 }
 
 // ---------------- LayerNorm (last-dim) ----------------

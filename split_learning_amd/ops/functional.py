@@ -141,13 +141,14 @@ class Conv2dFn(Function):
     to the pre-padded bounds-free gathers for A/B experiments."""
 
     @staticmethod
-    def forward(ctx, x, weight, bias, stride, padding):
+    def forward(ctx, x, weight, bias, stride, padding, bias_grad_zero):
         x = x.contiguous()
         y = native().conv2d_fwd(x, weight, bias, stride, padding)
         ctx.save_for_backward(x, weight)
         ctx.stride = stride
         ctx.padding = padding
         ctx.has_bias = bias is not None
+        ctx.bias_grad_zero = bias_grad_zero
         return y
 
     @staticmethod
@@ -162,12 +163,21 @@ class Conv2dFn(Function):
             gw = native().conv2d_bwd_weight(gy, x, weight.shape[2], weight.shape[3],
                                             ctx.stride, ctx.padding)
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            gb = native().conv2d_bwd_bias(gy)
-        return gx, gw, gb, None, None
+            if ctx.bias_grad_zero:
+                # conv feeding a TRAINING-mode BatchNorm: the bias gradient is
+                # analytically zero (sum over the batch of the BN-backward
+                # output is gamma*invstd*(sum_gy - sum_gy - sum_gy_xhat*
+                # sum(xhat)/n) and sum(xhat) == 0 by construction), so the
+                # two-kernel reduction is replaced by a constant
+                gb = gy.new_zeros(weight.shape[0])
+            else:
+                gb = native().conv2d_bwd_bias(gy)
+        return gx, gw, gb, None, None, None
 
 
-def conv2d(x, weight, bias=None, stride=1, padding=0):
-    return Conv2dFn.apply(x, weight, bias, int(stride), int(padding))
+def conv2d(x, weight, bias=None, stride=1, padding=0, bias_grad_zero=False):
+    return Conv2dFn.apply(x, weight, bias, int(stride), int(padding),
+                          bool(bias_grad_zero))
 
 
 # ---------------------------------------------------------------------------
@@ -177,12 +187,14 @@ def conv2d(x, weight, bias=None, stride=1, padding=0):
 class BatchNorm2dFn(Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training, momentum, eps,
-                fuse_relu):
+                fuse_relu, num_batches_tracked):
         x = x.contiguous()
         if training:
             # fused kernel: batch mean + biased-var invstd + torch-exact
-            # running-stat update (unbiased var), all device-side
+            # running-stat update (unbiased var) + num_batches_tracked++,
+            # all device-side in one launch pair
             mean, invstd = native().bn2d_stats_fused(x, running_mean, running_var,
+                                                     num_batches_tracked,
                                                      momentum, eps)
         else:
             mean = running_mean
@@ -197,20 +209,23 @@ class BatchNorm2dFn(Function):
     def backward(ctx, gy):
         x, gamma, mean, invstd, y = ctx.saved_tensors
         gy = gy.contiguous()
-        if ctx.fuse_relu:
-            gy = native().relu_bwd(gy, y)
+        # fused ReLU backward: the mask rides inside the BN backward kernels
+        # (no standalone relu_bwd launch, no extra full tensor pass)
+        relu_y = y if ctx.fuse_relu else None
         if ctx.training:
-            gx, ggamma, gbeta = native().bn2d_bwd(x, gy, gamma, mean, invstd)
+            gx, ggamma, gbeta = native().bn2d_bwd(x, gy, gamma, mean, invstd,
+                                                  relu_y)
         else:
             # eval-mode backward: per-channel affine with fixed stats
-            gx, ggamma, gbeta = native().bn2d_bwd_eval(x, gy, gamma, mean, invstd)
-        return gx, ggamma, gbeta, None, None, None, None, None, None
+            gx, ggamma, gbeta = native().bn2d_bwd_eval(x, gy, gamma, mean,
+                                                       invstd, relu_y)
+        return gx, ggamma, gbeta, None, None, None, None, None, None, None
 
 
 def batch_norm2d(x, gamma, beta, running_mean, running_var, training, momentum=0.1,
-                 eps=1e-5, fuse_relu=False):
+                 eps=1e-5, fuse_relu=False, num_batches_tracked=None):
     return BatchNorm2dFn.apply(x, gamma, beta, running_mean, running_var, training,
-                               momentum, eps, fuse_relu)
+                               momentum, eps, fuse_relu, num_batches_tracked)
 
 
 # ---------------------------------------------------------------------------

@@ -39,10 +39,11 @@ def _on_gpu(x: torch.Tensor) -> bool:
 
 
 class HipConv2d(nn.Conv2d):
-    def forward(self, x):
+    def forward(self, x, bias_grad_zero: bool = False):
         if _use_native("conv", x):
             return hf.conv2d(x, self.weight, self.bias,
-                             stride=self.stride[0], padding=self.padding[0])
+                             stride=self.stride[0], padding=self.padding[0],
+                             bias_grad_zero=bias_grad_zero)
         return super().forward(x)
 
 
@@ -56,11 +57,13 @@ class HipLinear(nn.Linear):
 class HipBatchNorm2d(nn.BatchNorm2d):
     def forward(self, x, fuse_relu: bool = False):
         if _use_native("bn", x):
-            if self.training and self.track_running_stats and self.num_batches_tracked is not None:
-                self.num_batches_tracked.add_(1)
+            # num_batches_tracked++ happens inside the stats finalize kernel
+            nbt = (self.num_batches_tracked
+                   if self.training and self.track_running_stats else None)
             return hf.batch_norm2d(x, self.weight, self.bias, self.running_mean,
                                    self.running_var, self.training, self.momentum,
-                                   self.eps, fuse_relu=fuse_relu)
+                                   self.eps, fuse_relu=fuse_relu,
+                                   num_batches_tracked=nbt)
         y = super().forward(x)
         return torch.relu(y) if fuse_relu else y
 
