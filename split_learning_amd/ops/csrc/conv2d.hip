@@ -369,8 +369,10 @@ static inline int slab_min_splits() {
   return v < 2 ? 2 : v;
 }
 
-// cap slab memory (splits * numel * 4B); above this stay with atomics
-static constexpr long SLK_SLAB_MAX_BYTES = 512l << 20;
+// cap slab memory (splits * numel * 4B); above this stay with atomics — the
+// reduce pass is traffic-proportional, and profiling shows big-output slabs
+// (tens of MB) cost more in reduce than their atomics would
+static constexpr long SLK_SLAB_MAX_BYTES = 32l << 20;
 
 // ---------------- padding / weight-flip prep kernels -----------------------
 
