@@ -82,6 +82,20 @@ class BertSdpaSelfAttention(nn.Module):
         return ctx.reshape(B, H, S, hd).permute(0, 2, 1, 3).reshape(B, S, E)
 
 
+def _dense_drop_res_ln(dense, dropout, ln, hidden_states, input_tensor,
+                       training):
+    """Shared sublayer epilogue: bias rides in the GEMM, then ONE fused
+    dropout+residual+LayerNorm kernel on GPU (norm.hip drop_res_ln_fwd —
+    SURVEY §2.4 fused bias-residual-LN row); stock ops on CPU."""
+    h = dense(hidden_states)
+    if h.is_cuda:
+        from ..ops.functional import dropout_residual_layer_norm
+        return dropout_residual_layer_norm(
+            h, input_tensor, ln.weight, ln.bias, eps=ln.eps,
+            p=dropout.p, training=training)
+    return ln(dropout(h) + input_tensor)
+
+
 class BertSelfOutput(nn.Module):
     def __init__(self, hidden_size, dropout_prob):
         super().__init__()
@@ -90,7 +104,8 @@ class BertSelfOutput(nn.Module):
         self.dropout = HipDropout(dropout_prob)
 
     def forward(self, hidden_states, input_tensor):
-        return self.LayerNorm(self.dropout(self.dense(hidden_states)) + input_tensor)
+        return _dense_drop_res_ln(self.dense, self.dropout, self.LayerNorm,
+                                  hidden_states, input_tensor, self.training)
 
 
 class BertAttention(nn.Module):
@@ -121,7 +136,8 @@ class BertOutput(nn.Module):
         self.dropout = HipDropout(dropout_prob)
 
     def forward(self, hidden_states, input_tensor):
-        return self.LayerNorm(self.dropout(self.dense(hidden_states)) + input_tensor)
+        return _dense_drop_res_ln(self.dense, self.dropout, self.LayerNorm,
+                                  hidden_states, input_tensor, self.training)
 
 
 class BertLayer(nn.Module):

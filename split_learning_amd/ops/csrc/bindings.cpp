@@ -28,6 +28,10 @@ std::vector<at::Tensor> bn2d_bwd(const at::Tensor&, const at::Tensor&,
 std::vector<at::Tensor> bn2d_bwd_eval(const at::Tensor&, const at::Tensor&,
                                       const at::Tensor&, const at::Tensor&,
                                       const at::Tensor&, c10::optional<at::Tensor>);
+std::vector<at::Tensor> drop_res_ln_fwd(const at::Tensor&, const at::Tensor&,
+                                        const at::Tensor&, const at::Tensor&,
+                                        double, double, int64_t,
+                                        c10::optional<at::Tensor>);
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor&, const at::Tensor&,
                                       const at::Tensor&, double);
 std::vector<at::Tensor> layernorm_bwd(const at::Tensor&, const at::Tensor&,
@@ -50,7 +54,8 @@ at::Tensor embedding_fwd(const at::Tensor&, const at::Tensor&);
 at::Tensor embedding_bwd(const at::Tensor&, const at::Tensor&, int64_t, int64_t);
 // attention.hip
 std::vector<at::Tensor> attn_fwd(const at::Tensor&, const at::Tensor&,
-                                 const at::Tensor&, double);
+                                 const at::Tensor&, double, double, int64_t,
+                                 c10::optional<at::Tensor>);
 // loss.hip
 at::Tensor softmax_fwd(const at::Tensor&);
 at::Tensor softmax_bwd(const at::Tensor&, const at::Tensor&);
@@ -94,6 +99,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn2d_bwd_eval", &slk::bn2d_bwd_eval, py::arg("x"), py::arg("gy"),
         py::arg("gamma"), py::arg("mean"), py::arg("invstd"),
         py::arg("relu_y") = py::none());
+  m.def("drop_res_ln_fwd", &slk::drop_res_ln_fwd, py::arg("x"), py::arg("res"),
+        py::arg("gamma"), py::arg("beta"), py::arg("eps"), py::arg("p") = 0.0,
+        py::arg("seed") = 0, py::arg("offset") = py::none());
   m.def("layernorm_fwd", &slk::layernorm_fwd);
   m.def("layernorm_bwd", &slk::layernorm_bwd);
   m.def("relu_fwd", &slk::relu_fwd);
@@ -109,7 +117,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2x2_bwd", &slk::maxpool2x2_bwd);
   m.def("embedding_fwd", &slk::embedding_fwd);
   m.def("embedding_bwd", &slk::embedding_bwd);
-  m.def("attn_fwd", &slk::attn_fwd);
+  m.def("attn_fwd", &slk::attn_fwd, py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("scale"), py::arg("p") = 0.0, py::arg("seed") = 0,
+        py::arg("offset") = py::none());
   m.def("softmax_fwd", &slk::softmax_fwd);
   m.def("softmax_bwd", &slk::softmax_bwd);
   m.def("ce_fwd", &slk::ce_fwd);

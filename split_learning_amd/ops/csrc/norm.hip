@@ -275,6 +275,100 @@ std::vector<at::Tensor> bn2d_bwd_eval(const at::Tensor& x, const at::Tensor& gy,
   return bn2d_bwd_impl(x, gy, gamma, mean, invstd, false, relu_y);
 }
 
+// ---------------- fused transformer epilogue ------------------------------
+// h = dropout(x) + residual;  y = LayerNorm(h)   in ONE kernel
+// (SURVEY §2.4 fused bias-residual-LN row: the bias itself rides in the
+// producing GEMM's epilogue, so the launches this removes are the dropout
+// and the aten residual add — the BertSelfOutput/Output chain goes
+// 4 launches -> 2 per sublayer).  h (the LN input) and the dropout mask are
+// written out for the backward, which composes the existing ln_bwd +
+// dropout_bwd + colsum kernels in Python.
+template <bool DROPOUT>
+__global__ void drop_res_ln_fwd_kernel(
+    const float* __restrict__ x, const float* __restrict__ res,
+    float* __restrict__ h, float* __restrict__ y,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ mean_out, float* __restrict__ invstd_out,
+    unsigned char* __restrict__ mask, int D, float eps, float p,
+    float inv_keep, uint64_t seed, const long* __restrict__ offset_ptr) {
+  __shared__ float scratch[16];
+  __shared__ float s_mean, s_invstd;
+  const uint64_t offset = offset_ptr ? (uint64_t)offset_ptr[0] : 0;
+  const long row = blockIdx.x;
+  const float* xr = x + row * D;
+  const float* rr = res + row * D;
+  float* hr = h + row * D;
+  float s = 0.f;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float v = xr[d];
+    if (DROPOUT) {
+      const bool keep = slk_uniform(seed, offset, row * D + d) >= p;
+      v = keep ? v * inv_keep : 0.f;
+      mask[row * D + d] = keep;
+    }
+    v += rr[d];
+    hr[d] = v;
+    s += v;
+  }
+  float total = slk_block_sum(s, scratch);
+  if (threadIdx.x == 0) s_mean = total / D;
+  __syncthreads();
+  const float m = s_mean;
+  float var = 0.f;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    const float t = hr[d] - m;
+    var += t * t;
+  }
+  float vtotal = slk_block_sum(var, scratch);
+  if (threadIdx.x == 0) s_invstd = rsqrtf(vtotal / D + eps);
+  __syncthreads();
+  const float is = s_invstd;
+  float* yr = y + row * D;
+  for (int d = threadIdx.x; d < D; d += blockDim.x)
+    yr[d] = (hr[d] - m) * is * gamma[d] + beta[d];
+  if (threadIdx.x == 0) {
+    mean_out[row] = m;
+    invstd_out[row] = is;
+  }
+}
+
+std::vector<at::Tensor> drop_res_ln_fwd(const at::Tensor& x,
+                                        const at::Tensor& res,
+                                        const at::Tensor& gamma,
+                                        const at::Tensor& beta, double eps,
+                                        double p, int64_t seed,
+                                        c10::optional<at::Tensor> offset) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.scalar_type() == at::kFloat);
+  const long R = x.size(0);
+  const int D = x.size(1);
+  auto h = at::empty_like(x);
+  auto y = at::empty_like(x);
+  auto mean = at::empty({R}, x.options());
+  auto invstd = at::empty({R}, x.options());
+  const bool drop = p > 0.0;
+  auto mask = at::empty({drop ? R : 0, D}, x.options().dtype(at::kByte));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int threads = D >= 256 ? 256 : 64;
+  if (drop) {
+    hipLaunchKernelGGL(drop_res_ln_fwd_kernel<true>, dim3(R), dim3(threads), 0,
+                       stream, x.data_ptr<float>(), res.data_ptr<float>(),
+                       h.data_ptr<float>(), y.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       mask.data_ptr<unsigned char>(), D, (float)eps, (float)p,
+                       (float)(1.0 / (1.0 - p)), (uint64_t)seed,
+                       offset.has_value() ? offset->data_ptr<long>() : nullptr);
+  } else {
+    hipLaunchKernelGGL(drop_res_ln_fwd_kernel<false>, dim3(R), dim3(threads), 0,
+                       stream, x.data_ptr<float>(), res.data_ptr<float>(),
+                       h.data_ptr<float>(), y.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       nullptr, D, (float)eps, 0.f, 1.f, 0, nullptr);
+  }
+  return {y, h, mean, invstd, mask};
+}
+
 // ---------------- LayerNorm (last-dim) ----------------
 
 // one block per row (rows up to ~4K cols; looping supports any D)

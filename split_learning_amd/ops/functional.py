@@ -378,37 +378,94 @@ def layer_norm(x, gamma, beta, eps=1e-5):
     return LayerNormFn.apply(x, gamma, beta, eps)
 
 
+class DropResLnFn(Function):
+    """Fused transformer epilogue y = LayerNorm(dropout(x) + residual) in one
+    kernel (norm.hip drop_res_ln_fwd; SURVEY §2.4 bias-residual-LN row — the
+    bias rides in the producing GEMM).  Backward composes the existing
+    layernorm_bwd + dropout_bwd kernels: the residual branch takes the LN
+    input-grad directly, the dense branch takes it through the mask."""
+
+    @staticmethod
+    def forward(ctx, x, res, gamma, beta, eps, p, seed, offset_t):
+        xs = x.reshape(-1, x.shape[-1]).contiguous()
+        rs = res.reshape(-1, res.shape[-1]).contiguous()
+        y, h, mean, invstd, mask = native().drop_res_ln_fwd(
+            xs, rs, gamma, beta, float(eps), float(p), seed, offset_t)
+        ctx.save_for_backward(h, gamma, mean, invstd, mask)
+        ctx.p = float(p)
+        ctx.x_shape = x.shape
+        return y.reshape(x.shape)
+
+    @staticmethod
+    def backward(ctx, gy):
+        h, gamma, mean, invstd, mask = ctx.saved_tensors
+        gys = gy.reshape(-1, gy.shape[-1]).contiguous()
+        gh, ggamma, gbeta = native().layernorm_bwd(gys, h, gamma, mean, invstd)
+        gres = gh.reshape(ctx.x_shape)
+        if ctx.p > 0.0:
+            gx = native().dropout_bwd(gh, mask, ctx.p).reshape(ctx.x_shape)
+        else:
+            gx = gres
+        return gx, gres, ggamma, gbeta, None, None, None, None
+
+
+def dropout_residual_layer_norm(x, res, gamma, beta, eps=1e-12, p=0.0,
+                                training=False):
+    pp = float(p) if training else 0.0
+    if pp > 0.0:
+        counter = _counter_for(x.device)
+        counter.add_(1)
+        seed, offset_t = _DROPOUT_STATE["seed"], counter
+    else:
+        seed, offset_t = 0, None
+    return DropResLnFn.apply(x, res, gamma, beta, eps, pp, seed, offset_t)
+
+
 # ---------------------------------------------------------------------------
 # Fused attention (small-S SDPA: S<=128, hd in {32,64} — the model zoo's regime)
 # ---------------------------------------------------------------------------
 
 class AttnFn(Function):
-    """softmax(Q@K^T * scale) @ V in ONE forward kernel (attention.hip): K/V
-    live in LDS, no scores round-trip through HBM, no separate scale/softmax
-    launches.  Backward reuses the GEMM + softmax-bwd kernels on the saved
-    probability matrix (identical math to the unfused path)."""
+    """dropout(softmax(Q@K^T * scale)) @ V in ONE forward kernel
+    (attention.hip): K/V live in LDS, no scores round-trip through HBM, no
+    separate scale/softmax/dropout launches.  Backward reuses the GEMM +
+    softmax-bwd + dropout-bwd kernels on the saved CLEAN probability matrix
+    and the kernel-produced mask (identical math to the unfused path)."""
 
     @staticmethod
-    def forward(ctx, q, k, v, scale):
-        out, probs = native().attn_fwd(q, k, v, scale)
-        ctx.save_for_backward(q, k, v, probs)
+    def forward(ctx, q, k, v, scale, p, seed, offset_t):
+        out, probs, mask = native().attn_fwd(q, k, v, scale, p, seed, offset_t)
+        ctx.save_for_backward(q, k, v, probs, mask)
         ctx.scale = scale
+        ctx.p = float(p)
         return out
 
     @staticmethod
     def backward(ctx, gy):
-        q, k, v, probs = ctx.saved_tensors
+        q, k, v, probs, mask = ctx.saved_tensors
         gy = gy.contiguous()
-        # dV = P^T @ gO ; dP = gO @ V^T ; dS = softmax_bwd(dP) * scale
-        gv = _matmul_raw(probs, gy, trans_a=True)
+        # pd = dropout(P); dV = pd^T @ gO ; dP = dropout_bwd(gO @ V^T) ;
+        # dS = softmax_bwd(dP, P) * scale
+        if ctx.p > 0.0:
+            mflat = mask.reshape(-1, mask.shape[-1])
+            pd = native().dropout_bwd(
+                probs.reshape(-1, probs.shape[-1]).contiguous(), mflat,
+                ctx.p).reshape(probs.shape)
+        else:
+            pd = probs
+        gv = _matmul_raw(pd, gy, trans_a=True)
         gp = _matmul_raw(gy, v, trans_b=True)
+        if ctx.p > 0.0:
+            gp = native().dropout_bwd(
+                gp.reshape(-1, gp.shape[-1]).contiguous(),
+                mask.reshape(-1, mask.shape[-1]), ctx.p).reshape(gp.shape)
         bhs = probs.shape[0] * probs.shape[1]
         gs = native().softmax_bwd(gp.reshape(bhs, -1).contiguous(),
                                   probs.reshape(bhs, -1)).reshape(probs.shape)
         gs = gs * ctx.scale
         gq = _matmul_raw(gs, k)
         gk = _matmul_raw(gs, q, trans_a=True)
-        return gq, gk, gv, None
+        return gq, gk, gv, None, None, None, None
 
 
 def attn_fused_ok(q) -> bool:
@@ -419,9 +476,18 @@ def attn_fused_ok(q) -> bool:
     return s <= 128 and hd <= 64 and (hd & (hd - 1)) == 0
 
 
-def attention(q, k, v, scale):
-    """Fused SDPA (no dropout variant); inputs [BH, S, hd] contiguous."""
-    return AttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(), scale)
+def attention(q, k, v, scale, dropout_p=0.0, training=False):
+    """Fused SDPA with optional in-kernel attention dropout;
+    inputs [BH, S, hd] contiguous."""
+    p = float(dropout_p) if training else 0.0
+    if p > 0.0:
+        counter = _counter_for(q.device)
+        counter.add_(1)
+        seed, offset_t = _DROPOUT_STATE["seed"], counter
+    else:
+        seed, offset_t = 0, None
+    return AttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(), scale,
+                        p, seed, offset_t)
 
 
 # ---------------------------------------------------------------------------

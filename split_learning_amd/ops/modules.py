@@ -136,13 +136,14 @@ class HipMultiheadAttention(nn.MultiheadAttention):
         qkv = hf.linear(x, self.in_proj_weight, self.in_proj_bias)  # [B,S,3E]
         q, k, v = qkv.split(E, dim=-1)
         scale = 1.0 / math.sqrt(hd)
-        no_drop = self.dropout == 0.0 or not self.training
-        if no_drop and S <= 128 and hd <= 64 and (hd & (hd - 1)) == 0 \
+        if S <= 128 and hd <= 64 and (hd & (hd - 1)) == 0 \
                 and "fattn" not in _DBG_TORCH:
             q = q.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd)
             k = k.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd)
             v = v.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd)
-            ctxv = hf.attention(q, k, v, scale)  # fused SDPA kernel
+            # fused SDPA kernel, in-kernel attention dropout (round 2)
+            ctxv = hf.attention(q, k, v, scale, dropout_p=self.dropout,
+                                training=self.training)
             ctxv = ctxv.reshape(B, H, S, hd).permute(0, 2, 1, 3).reshape(B, S, E)
         elif hf.use_lib_mm(B * H, S, S, hd):
             # library route (rocBLAS batched GEMM): keep [B,H,S,hd] strided
@@ -177,10 +178,11 @@ def attention_core(q, k, v, dropout_p: float = 0.0, training: bool = False,
     """scaled-dot-product attention on [B*H, S, hd] tensors via HIP kernels."""
     hd = q.shape[-1]
     s = scale if scale is not None else 1.0 / math.sqrt(hd)
-    if (_use_native("attn", q) and "fattn" not in _DBG_TORCH
-            and (dropout_p == 0.0 or not training) and q.dim() == 3
+    if (_use_native("attn", q) and "fattn" not in _DBG_TORCH and q.dim() == 3
             and q.shape[1] <= 128 and hd <= 64 and (hd & (hd - 1)) == 0):
-        return hf.attention(q, k, v, s)  # fused SDPA kernel
+        # fused SDPA kernel, attention dropout included (round-2: BERT's
+        # p=0.1 trains on the fused path)
+        return hf.attention(q, k, v, s, dropout_p=dropout_p, training=training)
     if _use_native("attn", q):
         scores = hf.matmul_f32(q, k, trans_b=True) * s
         probs = hf.softmax_lastdim(scores)
