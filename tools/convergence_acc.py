@@ -36,7 +36,7 @@ MOMENTUM = 0.5
 
 
 def make_data(n_train=4096, n_test=1024, seed=42, device="cuda",
-              sigma=4.0, label_noise=0.1):
+              sigma=3.0, label_noise=0.1):
     """Class-template task: x = template[y] + sigma*noise, 10 random
     templates.  (A random linear-projection rule was tried first and does
     NOT generalize for a CNN — all arms including stock torch sat at chance
