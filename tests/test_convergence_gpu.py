@@ -19,18 +19,21 @@ def test_split_learns_and_matches_monolithic():
     sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     from tools.convergence_acc import make_data, train_mono, train_split
 
-    steps = 500
+    # calibration (MI355X, gpurun_out/acc_*.log): 800 steps -> acc ~0.17-0.20,
+    # 1500 -> ~0.27, 5000 -> mono 0.569 / split 0.581 / stock-torch 0.531;
+    # 2000 steps lands mid-trajectory in a few seconds per arm
+    steps = 2000
     train, test = make_data(device="cuda")
     acc_mono, tr_mono = train_mono(train, test, steps, "cuda",
                                    log_every=steps // 5)
     acc_split, tr_split = train_split(train, test, steps, "cuda",
                                       log_every=steps // 5)
-    # learns: way above 10% chance
-    assert acc_mono > 0.45, f"monolithic failed to learn: {acc_mono}"
-    assert acc_split > 0.45, f"split failed to learn: {acc_split}"
+    # learns: far above 10% chance
+    assert acc_mono > 0.22, f"monolithic failed to learn: {acc_mono}"
+    assert acc_split > 0.22, f"split failed to learn: {acc_split}"
     # split == mono within run tolerance (same math at control-count 1;
     # kernel-order nondeterminism from atomics allows small drift)
     assert abs(acc_mono - acc_split) < 0.08, (acc_mono, acc_split)
     # loss decreased materially in both arms
-    assert tr_mono[-1] < tr_mono[0] * 0.7, tr_mono
-    assert tr_split[-1] < tr_split[0] * 0.7, tr_split
+    assert tr_mono[-1] < tr_mono[0] * 0.75, tr_mono
+    assert tr_split[-1] < tr_split[0] * 0.75, tr_split
