@@ -137,7 +137,8 @@ class HipMultiheadAttention(nn.MultiheadAttention):
         q, k, v = qkv.split(E, dim=-1)
         scale = 1.0 / math.sqrt(hd)
         if S <= 128 and hd <= 64 and (hd & (hd - 1)) == 0 \
-                and "fattn" not in _DBG_TORCH:
+                and "fattn" not in _DBG_TORCH \
+                and _fused_attn_worth_it(B * H, S, self.dropout, self.training):
             q = q.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd)
             k = k.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd)
             v = v.reshape(B, S, H, hd).permute(0, 2, 1, 3).reshape(B * H, S, hd)
@@ -173,15 +174,27 @@ class HipMultiheadAttention(nn.MultiheadAttention):
         return out, None
 
 
+def _fused_attn_worth_it(bh, s_len, dropout_p, training) -> bool:
+    """The fused kernel wins when launches dominate; with in-kernel dropout
+    its per-element RNG (two splitmix64 hashes) makes it VALU-bound, and at
+    BERT scale (BH=384, S=128 -> 6.3M mask draws) the routed GEMM path
+    measured ~10x faster (profiles/SUMMARY.md round 2).  Dropout-off: always
+    fused.  Dropout-on: fused only for small score matrices."""
+    if dropout_p == 0.0 or not training:
+        return True
+    return bh * s_len * s_len <= (1 << 21)
+
+
 def attention_core(q, k, v, dropout_p: float = 0.0, training: bool = False,
                    scale: Optional[float] = None):
     """scaled-dot-product attention on [B*H, S, hd] tensors via HIP kernels."""
     hd = q.shape[-1]
     s = scale if scale is not None else 1.0 / math.sqrt(hd)
     if (_use_native("attn", q) and "fattn" not in _DBG_TORCH and q.dim() == 3
-            and q.shape[1] <= 128 and hd <= 64 and (hd & (hd - 1)) == 0):
-        # fused SDPA kernel, attention dropout included (round-2: BERT's
-        # p=0.1 trains on the fused path)
+            and q.shape[1] <= 128 and hd <= 64 and (hd & (hd - 1)) == 0
+            and _fused_attn_worth_it(q.shape[0], q.shape[1], dropout_p,
+                                     training)):
+        # fused SDPA kernel, attention dropout included in-kernel
         return hf.attention(q, k, v, s, dropout_p=dropout_p, training=training)
     if _use_native("attn", q):
         scores = hf.matmul_f32(q, k, trans_b=True) * s
