@@ -75,6 +75,14 @@ DEFAULT_CONFIG: Dict[str, Any] = {
         "kind": "loopback",
         "master-addr": "127.0.0.1",
         "master-port": 29571,
+        # FedAvg path: "control" ships state dicts through the control plane
+        # (reference semantics); "rccl" all-reduces weight*size over the
+        # stage-group communicator (sequential policies force "control")
+        "fedavg": "control",
+        # START parameter delivery: "auto" broadcasts the full model over
+        # RCCL when safe (concurrent policy, all ranks accepted, dist live),
+        # "control" always ships per-client slices, "rccl" forces broadcast
+        "params": "auto",
     },
     "device": {
         "dtype": "float32",

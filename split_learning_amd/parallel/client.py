@@ -129,6 +129,17 @@ class ClientRuntime:
         self.data_name = msg["data_name"]
         self.learning = msg["learning"]
         layers = msg["layers"]
+        if msg.get("param_bcast"):
+            # parameters arrive as ONE RCCL broadcast of the full model over
+            # xGMI (bcast.py); rank 0's client feeds the server-staged state
+            # into the collective, every rank slices its partition locally
+            from . import bcast
+            import torch.distributed as dist
+            staged = bcast.take_staged("round") if dist.get_rank() == 0 else None
+            full = bcast.broadcast_full_state(self.model_name, self.data_name,
+                                              staged, self.device)
+            part = build_partition(self.model_name, self.data_name, layers)
+            state_dict = {k: full[k] for k in part.state_dict().keys()}
         self.cluster = msg.get("cluster", self.cluster) or 0
         self.n_stages = msg["n_stages"]
         refresh = msg.get("refresh", True)

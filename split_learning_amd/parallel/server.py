@@ -331,12 +331,31 @@ class Server:
                     groups.append(ids)
         return groups
 
-    def _send_start(self, rec, full_state, state_override=None):
+    def _param_bcast_ok(self, active, full_state) -> bool:
+        """START parameters go out as ONE RCCL broadcast of the full model
+        (bcast.py) instead of per-client control-plane blobs when: the
+        transport allows it, the policy is concurrent (every rank reaches the
+        collective this round), torch.distributed is live in this process
+        (p2p mode: the server thread shares rank 0's process), and every rank
+        is an accepted client (a rejected rank would never join)."""
+        import torch.distributed as dist
+        mode = (self.config.get("transport") or {}).get("params", "auto")
+        if mode not in ("auto", "rccl") or not self.RCCL_FEDAVG_OK:
+            return False
+        if full_state is None or not (dist.is_available() and dist.is_initialized()):
+            return False
+        ids = sorted(c["client_id"] for c in active)
+        return ids == list(range(dist.get_world_size()))
+
+    def _send_start(self, rec, full_state, state_override=None,
+                    param_bcast=False):
         layers = self._stage_layers(rec)
-        state = state_override if state_override is not None             else self._slice_state(full_state, layers)
+        state = None if param_bcast else (
+            state_override if state_override is not None
+            else self._slice_state(full_state, layers))
         self.control.send(f"client_{rec['client_id']}", {
             "action": "START", "message": "Server accept the connection!",
-            "parameters": state, "layers": layers,
+            "parameters": state, "param_bcast": param_bcast, "layers": layers,
             "model_name": self.model_name, "data_name": self.data_name,
             "learning": self.learning, "label_count": rec["label"],
             "refresh": self.refresh, "cluster": rec["cluster"],
@@ -378,6 +397,10 @@ class Server:
     def notify_clients(self, start: bool = True):
         full_state = self._load_ckpt() if start else None
         active = [c for c in self.list_clients if c["train"]]
+        use_bcast = start and self._param_bcast_ok(active, full_state)
+        if use_bcast:
+            from . import bcast
+            bcast.stage_full_state("round", full_state)
         for rec in self.list_clients:
             if not start:
                 self._send_stop(rec)
@@ -386,7 +409,7 @@ class Server:
                 if not self.reject:
                     self._send_stop(rec, "Reject Device")
                 continue
-            self._send_start(rec, full_state)
+            self._send_start(rec, full_state, param_bcast=use_bcast)
         if not start:
             self.stopped = True
             return
