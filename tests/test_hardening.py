@@ -88,9 +88,9 @@ def test_save_only_round_progress(tmp_path):
     from split_learning_amd.parallel import server as server_mod
     orig = server_mod.Server._send_start
 
-    def spy(self, rec, full_state, state_override=None):
+    def spy(self, rec, full_state, state_override=None, **kw):
         seen_params.append(full_state is not None or state_override is not None)
-        return orig(self, rec, full_state, state_override)
+        return orig(self, rec, full_state, state_override, **kw)
 
     server_mod.Server._send_start = spy
     try:
