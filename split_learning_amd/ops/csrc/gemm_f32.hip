@@ -26,23 +26,28 @@ struct StridedGather {
   long sAb, sAm, sAk;
   long sBb, sBk, sBn;
 
+  // Contexts hold 32-bit OFFSETS against the functor's uniform (SGPR) base
+  // pointers rather than per-context 64-bit pointers: the big-tile kernel
+  // keeps 8 A-contexts live and the pointer form cost 16 VGPR of its
+  // register budget (the 128x128 kernel sat at 3 waves/SIMD; see
+  // tile_gemm.h).  Host guards numel < 2^31 (matmul_f32).
   struct KCtx {};
-  struct ACtx { const float* p; bool valid; };
-  struct BCtx { const float* p; bool valid; };
+  struct ACtx { int off; bool valid; };
+  struct BCtx { int off; bool valid; };
 
   __device__ KCtx prepK(int) const { return {}; }
   __device__ ACtx prepA(int b, int m, bool valid, int) const {
-    return {A + (long)b * sAb + (long)m * sAm, valid};
+    return {(int)((long)b * sAb + (long)m * sAm), valid};
   }
   __device__ float loadA(const ACtx& c, const KCtx&, int k, bool kv) const {
-    const float v = c.p[(long)k * sAk];
+    const float v = A[c.off + (long)k * sAk];
     return (c.valid & kv) ? v : 0.f;
   }
   __device__ BCtx prepB(int b, int n, bool valid, int) const {
-    return {B + (long)b * sBb + (long)n * sBn, valid};
+    return {(int)((long)b * sBb + (long)n * sBn), valid};
   }
   __device__ float loadB(const BCtx& c, const KCtx&, int k, bool kv) const {
-    const float v = c.p[(long)k * sBk];
+    const float v = B[c.off + (long)k * sBk];
     return (c.valid & kv) ? v : 0.f;
   }
 };
@@ -101,6 +106,8 @@ at::Tensor matmul_f32(const at::Tensor& a, const at::Tensor& b, bool ta, bool tb
               "matmul_f32 is fp32-only");
   TORCH_CHECK(a.dim() == b.dim() && (a.dim() == 2 || a.dim() == 3),
               "matmul_f32: 2-D or 3-D tensors, same rank");
+  TORCH_CHECK(a.numel() < (1ll << 31) && b.numel() < (1ll << 31),
+              "matmul_f32: int32 gather offsets (tensor too large)");
   const int M = ta ? a.size(-1) : a.size(-2);
   const int Ka = ta ? a.size(-2) : a.size(-1);
   const int Kb = tb ? b.size(-1) : b.size(-2);

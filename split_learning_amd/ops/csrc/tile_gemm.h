@@ -275,13 +275,26 @@ __device__ __forceinline__ int slk_big_perm(int c) {
   return 2 * (c & 31) + ((c >> 5) & 1) + 64 * (c >> 6);
 }
 
+// 2nd launch-bounds arg = min waves per SIMD: forces the allocator to 128
+// total registers (64 VGPR + the 64 MFMA accumulator AGPRs) so FOUR waves
+// fit per SIMD instead of three — the allocator does not get there on its
+// own (74 VGPR unforced)
 template <typename Gather, typename Store>
-__global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
-    Gather g, Store st, int M, int N, int K, int split_k, int k_per_split) {
-  __shared__ __align__(16) float ldsA[SLK_BK * SLK_LDS_ROW2];
-  __shared__ __align__(16) float ldsB[SLK_BK * SLK_LDS_ROW2];
+__global__ __launch_bounds__(256, 4) void slk_mfma_gemm_kernel_big(
+    Gather g, Store st, int M, int N, int K, int split_k, int k_per_split,
+    int xcd_remap) {
+  // (LDS ping-pong was tried here too — 2 x 16.6 KB keeps 4 blocks/CU so it
+  // is occupancy-free — and still measured ~3% SLOWER at 4096^3, likely the
+  // +3 VGPR spills in the forced-128-register budget.  Two-barrier loop
+  // retained; profiles/SUMMARY.md round 2.)
+  __shared__ __align__(16) float ldsA[1][SLK_BK * SLK_LDS_ROW2];
+  __shared__ __align__(16) float ldsB[1][SLK_BK * SLK_LDS_ROW2];
 
-  const int tile_n = blockIdx.x;
+  // XCD-aware contiguous n-tile ownership (see the 64x64 kernel's note)
+  int tile_n = blockIdx.x;
+  if (xcd_remap && (gridDim.x & 7) == 0) {
+    tile_n = (blockIdx.x & 7) * (gridDim.x >> 3) + (blockIdx.x >> 3);
+  }
   const int tile_m = blockIdx.y;
   const int batch = blockIdx.z / split_k;
   const int ks = blockIdx.z % split_k;
@@ -332,18 +345,22 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
     }
   };
 
-  load_tile(k_begin);
-  for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
+  auto stage_to = [&](int p) {
     #pragma unroll
     for (int i = 0; i < RA; ++i) {
       const int idx = tid + i * 256;
-      ldsA[(idx & 15) * SLK_LDS_ROW2 + slk_big_perm(idx >> 4)] = ra[i];
+      ldsA[p][(idx & 15) * SLK_LDS_ROW2 + slk_big_perm(idx >> 4)] = ra[i];
     }
     #pragma unroll
     for (int i = 0; i < RB; ++i) {
       const int idx = tid + i * 256;
-      ldsB[(idx >> 7) * SLK_LDS_ROW2 + slk_big_perm(idx & 127)] = rb[i];
+      ldsB[p][(idx >> 7) * SLK_LDS_ROW2 + slk_big_perm(idx & 127)] = rb[i];
     }
+  };
+
+  load_tile(k_begin);
+  for (int k0 = k_begin; k0 < k_end; k0 += SLK_BK) {
+    stage_to(0);
     __syncthreads();
     if (k0 + SLK_BK < k_end) load_tile(k0 + SLK_BK);
 
@@ -352,9 +369,9 @@ __global__ __launch_bounds__(256) void slk_mfma_gemm_kernel_big(
       const int kr = kk * 2 + kh;
       // pairs (wm+l31, wm+32+l31) / (wn+l31, wn+32+l31) are adjacent dwords
       const f32x2 a01 = *reinterpret_cast<const f32x2*>(
-          &ldsA[kr * SLK_LDS_ROW2 + 64 * (wm >> 6) + 2 * l31]);
+          &ldsA[0][kr * SLK_LDS_ROW2 + 64 * (wm >> 6) + 2 * l31]);
       const f32x2 b01 = *reinterpret_cast<const f32x2*>(
-          &ldsB[kr * SLK_LDS_ROW2 + 64 * (wn >> 6) + 2 * l31]);
+          &ldsB[0][kr * SLK_LDS_ROW2 + 64 * (wn >> 6) + 2 * l31]);
       acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.x, b01.x, acc[0][0], 0, 0, 0);
       acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.x, b01.y, acc[0][1], 0, 0, 0);
       acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a01.y, b01.x, acc[1][0], 0, 0, 0);
@@ -419,17 +436,17 @@ inline void slk_launch_gemm(const Gather& g, const Store& st, int M, int N, int 
   // round k_per_split up to a BK multiple so every split starts aligned
   k_per_split = ((k_per_split + SLK_BK - 1) / SLK_BK) * SLK_BK;
   split_k = ceil_div(K, k_per_split);
-  if (big) {
-    dim3 grid(ceil_div(N, SLK_BN2), ceil_div(M, SLK_BM2), n_batch * split_k);
-    hipLaunchKernelGGL((slk_mfma_gemm_kernel_big<Gather, Store>), grid, dim3(256),
-                       0, stream, g, st, M, N, K, split_k, k_per_split);
-    return;
-  }
-  dim3 grid(ceil_div(N, SLK_BN), ceil_div(M, SLK_BM), n_batch * split_k);
   static const int xcd = [] {
     const char* e = std::getenv("SLK_XCD");
     return e ? atoi(e) : 1;
   }();
+  if (big) {
+    dim3 grid(ceil_div(N, SLK_BN2), ceil_div(M, SLK_BM2), n_batch * split_k);
+    hipLaunchKernelGGL((slk_mfma_gemm_kernel_big<Gather, Store>), grid, dim3(256),
+                       0, stream, g, st, M, N, K, split_k, k_per_split, xcd);
+    return;
+  }
+  dim3 grid(ceil_div(N, SLK_BN), ceil_div(M, SLK_BM), n_batch * split_k);
   if (slk_pp_mode()) {
     hipLaunchKernelGGL((slk_mfma_gemm_kernel<Gather, Store, true>), grid,
                        dim3(256), 0, stream, g, st, M, N, K, split_k,
