@@ -359,12 +359,13 @@ static void slab_reduce(const at::Tensor& slab, int splits, at::Tensor& out,
                      numel);
 }
 
-// slab mode threshold: slab stores beat atomics once several splits pile on
-// one output (env-tunable for A/B sweeps)
+// slab mode threshold: slab stores beat atomics once MANY splits pile on
+// one output; at mild splits (<8) the reduce pass's launch+ramp overhead
+// loses to cheap atomics (bench sweep, profiles/SUMMARY.md round 2)
 static inline int slab_min_splits() {
   static int v = [] {
     const char* e = std::getenv("SLK_SLAB_MIN");
-    return e ? atoi(e) : 4;
+    return e ? atoi(e) : 8;
   }();
   return v < 2 ? 2 : v;
 }
