@@ -22,6 +22,7 @@ BUILD_DIR = os.path.join(HERE, "_build")
 SOURCES = [
     "gemm_f32.hip",
     "conv2d.hip",
+    "wino.hip",
     "norm.hip",
     "elementwise.hip",
     "loss.hip",

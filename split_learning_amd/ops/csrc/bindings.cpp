@@ -7,6 +7,9 @@ at::Tensor matmul_f32(const at::Tensor&, const at::Tensor&, bool, bool,
                       c10::optional<at::Tensor>, bool);
 at::Tensor linear_fwd(const at::Tensor&, const at::Tensor&, c10::optional<at::Tensor>);
 at::Tensor colsum_f32(const at::Tensor&);
+// wino.hip
+at::Tensor conv2d_wino(const at::Tensor&, const at::Tensor&,
+                       c10::optional<at::Tensor>, int, bool);
 // conv2d.hip
 at::Tensor pad_nchw(const at::Tensor&, int);
 at::Tensor conv2d_fwd(const at::Tensor&, const at::Tensor&, c10::optional<at::Tensor>,
@@ -82,6 +85,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out") = py::none(), py::arg("accumulate") = false);
   m.def("linear_fwd", &slk::linear_fwd);
   m.def("colsum_f32", &slk::colsum_f32);
+  m.def("conv2d_wino", &slk::conv2d_wino, py::arg("x"), py::arg("w"),
+        py::arg("bias") = py::none(), py::arg("pad") = 1,
+        py::arg("flip") = false);
   m.def("pad_nchw", &slk::pad_nchw);
   m.def("conv2d_fwd", &slk::conv2d_fwd, py::arg("x"), py::arg("w"),
         py::arg("bias"), py::arg("stride"), py::arg("pad"),

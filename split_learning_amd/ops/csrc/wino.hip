@@ -1,0 +1,227 @@
+// Winograd F(2x2, 3x3) convolution for stride-1 3x3 convs (fp32).
+//
+// y = A^T [ (G g G^T) .* (B^T d B) ] A   per 2x2 output tile (4x4 input
+// patch, overlap 2).  2.25x fewer MACs than direct: the elementwise product
+// becomes 16 frequency-batched GEMMs  M[f] = U[f][Co,Ci] @ V[f][Ci,T]
+// (T = B * OH/2 * OW/2) which run on the existing MFMA tile framework as ONE
+// batched launch.  Unfused (transforms round-trip V/M through HBM), so it
+// wins where the GEMM is compute-dominated — the high-Ci tail of VGG16 —
+// and loses to the direct implicit-GEMM kernel on the big-spatial low-Ci
+// layers where the 4x data inflation of V/M eats the MAC savings
+// (measured routing in conv2d.hip; MIOpen's advantage on those layers is a
+// FUSED Winograd with in-kernel transforms).
+//
+// Backward-data reuses the whole machinery: gx = winograd-conv of gy with
+// the 180-degree-rotated, Co/Ci-transposed weights at pad' = KH-1-p.
+// Backward-weight stays on the direct gather (a correlation, not a 3x3
+// conv over the same axes).
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.h"
+
+namespace slk {
+
+// weight transform U[f][Co][Ci] = (G g G^T)[f]; FLIP transposes Co/Ci and
+// rotates the taps 180 degrees (the bwd-data weight)
+template <bool FLIP>
+__global__ void wino_wt_kernel(const float* __restrict__ w,
+                               float* __restrict__ U, int Co, int Ci) {
+  const long total = (long)Co * Ci;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    // output index: U[f][r][c] with (r, c) = FLIP ? (ci, co) : (co, ci)
+    const int c = (int)(i % (FLIP ? Co : Ci));
+    const int r = (int)(i / (FLIP ? Co : Ci));
+    const int co = FLIP ? c : r;
+    const int ci = FLIP ? r : c;
+    float g[3][3];
+    #pragma unroll
+    for (int a = 0; a < 3; ++a)
+      #pragma unroll
+      for (int b = 0; b < 3; ++b)
+        g[a][b] = w[(((long)co * (FLIP ? Ci : Ci) + ci) * 3 +
+                     (FLIP ? 2 - a : a)) * 3 + (FLIP ? 2 - b : b)];
+    // Gg: [4][3],  G = [[1,0,0],[.5,.5,.5],[.5,-.5,.5],[0,0,1]]
+    float t[4][3];
+    #pragma unroll
+    for (int b = 0; b < 3; ++b) {
+      t[0][b] = g[0][b];
+      t[1][b] = 0.5f * (g[0][b] + g[1][b] + g[2][b]);
+      t[2][b] = 0.5f * (g[0][b] - g[1][b] + g[2][b]);
+      t[3][b] = g[2][b];
+    }
+    #pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      const float u0 = t[a][0];
+      const float u1 = 0.5f * (t[a][0] + t[a][1] + t[a][2]);
+      const float u2 = 0.5f * (t[a][0] - t[a][1] + t[a][2]);
+      const float u3 = t[a][2];
+      U[((long)(a * 4 + 0) * total) + i] = u0;
+      U[((long)(a * 4 + 1) * total) + i] = u1;
+      U[((long)(a * 4 + 2) * total) + i] = u2;
+      U[((long)(a * 4 + 3) * total) + i] = u3;
+    }
+  }
+}
+
+// input transform V[f][Ci][T], T = B*tH*tW, tile t = (b, th, tw); the 4x4
+// patch starts at (th*2 - pad, tw*2 - pad), zero-padded at the borders
+__global__ void wino_in_kernel(const float* __restrict__ x,
+                               float* __restrict__ V, int B, int Ci, int H,
+                               int W, int tH, int tW, int pad, FastDiv d_T,
+                               FastDiv d_thw, FastDiv d_tw) {
+  const int T = B * tH * tW;
+  const long total = (long)Ci * T;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const unsigned cc = d_T.div((unsigned)i);
+    const unsigned t = d_T.mod((unsigned)i, cc);
+    const unsigned b = d_thw.div(t);
+    const unsigned rem = d_thw.mod(t, b);
+    const unsigned th = d_tw.div(rem);
+    const unsigned tw = d_tw.mod(rem, th);
+    const int ih0 = (int)th * 2 - pad;
+    const int iw0 = (int)tw * 2 - pad;
+    const float* xp = x + ((long)b * Ci + cc) * H * W;
+    float d[4][4];
+    #pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      const int ih = ih0 + a;
+      const bool hv = (unsigned)ih < (unsigned)H;
+      #pragma unroll
+      for (int bb = 0; bb < 4; ++bb) {
+        const int iw = iw0 + bb;
+        const bool v = hv && (unsigned)iw < (unsigned)W;
+        d[a][bb] = v ? xp[(long)ih * W + iw] : 0.f;
+      }
+    }
+    // B^T d: rows [d0-d2, d1+d2, d2-d1, d1-d3]
+    float u[4][4];
+    #pragma unroll
+    for (int bb = 0; bb < 4; ++bb) {
+      u[0][bb] = d[0][bb] - d[2][bb];
+      u[1][bb] = d[1][bb] + d[2][bb];
+      u[2][bb] = d[2][bb] - d[1][bb];
+      u[3][bb] = d[1][bb] - d[3][bb];
+    }
+    #pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      const float v0 = u[a][0] - u[a][2];
+      const float v1 = u[a][1] + u[a][2];
+      const float v2 = u[a][2] - u[a][1];
+      const float v3 = u[a][1] - u[a][3];
+      V[((long)(a * 4 + 0) * Ci + cc) * T + t] = v0;
+      V[((long)(a * 4 + 1) * Ci + cc) * T + t] = v1;
+      V[((long)(a * 4 + 2) * Ci + cc) * T + t] = v2;
+      V[((long)(a * 4 + 3) * Ci + cc) * T + t] = v3;
+    }
+  }
+}
+
+// output transform y[b][co][2th..][2tw..] = A^T M A (+bias)
+__global__ void wino_out_kernel(const float* __restrict__ Mm,
+                                float* __restrict__ y,
+                                const float* __restrict__ bias, int B, int Co,
+                                int OH, int OW, int tH, int tW, FastDiv d_T,
+                                FastDiv d_thw, FastDiv d_tw) {
+  const int T = B * tH * tW;
+  const long total = (long)Co * T;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const unsigned co = d_T.div((unsigned)i);
+    const unsigned t = d_T.mod((unsigned)i, co);
+    const unsigned b = d_thw.div(t);
+    const unsigned rem = d_thw.mod(t, b);
+    const unsigned th = d_tw.div(rem);
+    const unsigned tw = d_tw.mod(rem, th);
+    float m[4][4];
+    #pragma unroll
+    for (int f = 0; f < 16; ++f)
+      m[f >> 2][f & 3] = Mm[((long)f * Co + co) * T + t];
+    // A^T m: [m0+m1+m2, m1-m2-m3]
+    float u[2][4];
+    #pragma unroll
+    for (int bb = 0; bb < 4; ++bb) {
+      u[0][bb] = m[0][bb] + m[1][bb] + m[2][bb];
+      u[1][bb] = m[1][bb] - m[2][bb] - m[3][bb];
+    }
+    const float bv = bias != nullptr ? bias[co] : 0.f;
+    float* yp = y + ((long)b * Co + co) * OH * OW + (long)th * 2 * OW + tw * 2;
+    yp[0] = u[0][0] + u[0][1] + u[0][2] + bv;
+    yp[1] = u[0][1] - u[0][2] - u[0][3] + bv;
+    yp[OW] = u[1][0] + u[1][1] + u[1][2] + bv;
+    yp[OW + 1] = u[1][1] - u[1][2] - u[1][3] + bv;
+  }
+}
+
+// host entry: 3x3 stride-1 conv via F(2x2,3x3); flip=true computes the
+// bwd-data conv (weights rotated + Co/Ci transposed, pad' = 2 - pad)
+at::Tensor matmul_f32(const at::Tensor&, const at::Tensor&, bool, bool,
+                      c10::optional<at::Tensor>, bool);
+
+at::Tensor conv2d_wino(const at::Tensor& x, const at::Tensor& w,
+                       c10::optional<at::Tensor> bias, int pad, bool flip) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.scalar_type() == at::kFloat);
+  TORCH_CHECK(w.size(2) == 3 && w.size(3) == 3, "wino: 3x3 only");
+  auto xc = x.contiguous();
+  auto wc = w.contiguous();
+  const int B = x.size(0), H = x.size(2), W = x.size(3);
+  const int Co = flip ? w.size(1) : w.size(0);
+  const int Ci = flip ? w.size(0) : w.size(1);
+  TORCH_CHECK(x.size(1) == Ci, "wino: channel mismatch");
+  const int OH = H + 2 * pad - 2, OW = W + 2 * pad - 2;
+  TORCH_CHECK(OH % 2 == 0 && OW % 2 == 0, "wino: even output dims only");
+  const int tH = OH / 2, tW = OW / 2;
+  const int T = B * tH * tW;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+
+  auto U = at::empty({16, Co, Ci}, x.options());
+  auto V = at::empty({16, Ci, T}, x.options());
+  {
+    const long tot = (long)Co * Ci;
+    const int grid = (int)std::min<long>((tot + 255) / 256, 4096);
+    if (flip) {
+      hipLaunchKernelGGL(wino_wt_kernel<true>, dim3(grid), dim3(256), 0,
+                         stream, wc.data_ptr<float>(), U.data_ptr<float>(),
+                         Co, Ci);
+    } else {
+      hipLaunchKernelGGL(wino_wt_kernel<false>, dim3(grid), dim3(256), 0,
+                         stream, wc.data_ptr<float>(), U.data_ptr<float>(),
+                         Co, Ci);
+    }
+  }
+  {
+    FastDiv d_T, d_thw, d_tw;
+    d_T.init(T);
+    d_thw.init(tH * tW);
+    d_tw.init(tW);
+    const long tot = (long)Ci * T;
+    const int grid = (int)std::min<long>((tot + 255) / 256, 8192);
+    hipLaunchKernelGGL(wino_in_kernel, dim3(grid), dim3(256), 0, stream,
+                       xc.data_ptr<float>(), V.data_ptr<float>(), B, Ci, H, W,
+                       tH, tW, pad, d_T, d_thw, d_tw);
+  }
+  // M[f] = U[f] @ V[f]: one batched MFMA GEMM launch (batch = 16)
+  auto Mm = matmul_f32(U, V, false, false, c10::nullopt, false);
+  auto y = at::empty({B, Co, OH, OW}, x.options());
+  {
+    FastDiv d_T, d_thw, d_tw;
+    d_T.init(T);
+    d_thw.init(tH * tW);
+    d_tw.init(tW);
+    const long tot = (long)Co * T;
+    const int grid = (int)std::min<long>((tot + 255) / 256, 8192);
+    hipLaunchKernelGGL(wino_out_kernel, dim3(grid), dim3(256), 0, stream,
+                       Mm.data_ptr<float>(), y.data_ptr<float>(),
+                       bias.has_value() ? bias->data_ptr<float>() : nullptr,
+                       B, Co, OH, OW, tH, tW, d_T, d_thw, d_tw);
+  }
+  return y;
+}
+
+}  // namespace slk
