@@ -23,26 +23,27 @@
 
 namespace slk {
 
-// weight transform U[f][Co][Ci] = (G g G^T)[f]; FLIP transposes Co/Ci and
-// rotates the taps 180 degrees (the bwd-data weight)
+// weight transform U[f] = (G g G^T)[f] over the weight's true dims
+// (Cw0, Cw1) = (w.size(0), w.size(1)); FLIP writes U[f][Cw1][Cw0] with the
+// taps rotated 180 degrees (the bwd-data weight), else U[f][Cw0][Cw1]
 template <bool FLIP>
 __global__ void wino_wt_kernel(const float* __restrict__ w,
-                               float* __restrict__ U, int Co, int Ci) {
-  const long total = (long)Co * Ci;
+                               float* __restrict__ U, int Cw0, int Cw1) {
+  const long total = (long)Cw0 * Cw1;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
-    // output index: U[f][r][c] with (r, c) = FLIP ? (ci, co) : (co, ci)
-    const int c = (int)(i % (FLIP ? Co : Ci));
-    const int r = (int)(i / (FLIP ? Co : Ci));
-    const int co = FLIP ? c : r;
-    const int ci = FLIP ? r : c;
+    const int cols = FLIP ? Cw0 : Cw1;
+    const int c = (int)(i % cols);
+    const int r = (int)(i / cols);
+    const int co = FLIP ? c : r;   // index into w's dim 0
+    const int ci = FLIP ? r : c;   // index into w's dim 1
     float g[3][3];
     #pragma unroll
     for (int a = 0; a < 3; ++a)
       #pragma unroll
       for (int b = 0; b < 3; ++b)
-        g[a][b] = w[(((long)co * (FLIP ? Ci : Ci) + ci) * 3 +
+        g[a][b] = w[(((long)co * Cw1 + ci) * 3 +
                      (FLIP ? 2 - a : a)) * 3 + (FLIP ? 2 - b : b)];
     // Gg: [4][3],  G = [[1,0,0],[.5,.5,.5],[.5,-.5,.5],[0,0,1]]
     float t[4][3];
@@ -185,14 +186,15 @@ at::Tensor conv2d_wino(const at::Tensor& x, const at::Tensor& w,
   {
     const long tot = (long)Co * Ci;
     const int grid = (int)std::min<long>((tot + 255) / 256, 4096);
+    // always the WEIGHT's own dims; FLIP handles the transpose internally
     if (flip) {
       hipLaunchKernelGGL(wino_wt_kernel<true>, dim3(grid), dim3(256), 0,
                          stream, wc.data_ptr<float>(), U.data_ptr<float>(),
-                         Co, Ci);
+                         (int)w.size(0), (int)w.size(1));
     } else {
       hipLaunchKernelGGL(wino_wt_kernel<false>, dim3(grid), dim3(256), 0,
                          stream, wc.data_ptr<float>(), U.data_ptr<float>(),
-                         Co, Ci);
+                         (int)w.size(0), (int)w.size(1));
     }
   }
   {

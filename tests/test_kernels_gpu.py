@@ -565,3 +565,27 @@ def test_drop_res_ln_fused():
     yref = F.layer_norm(href, (D,), gamma.detach(), beta.detach(), eps=1e-12)
     assert_close(y2, yref, atol=1e-4, rtol=1e-4, what="drl dropout fwd")
     assert_close(h2, href, atol=1e-5, rtol=1e-5, what="drl hidden")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("ci,hw,co", [(64, 32, 64), (64, 16, 128),
+                                      (128, 8, 256), (512, 4, 512)])
+def test_conv2d_winograd(ci, hw, co):
+    """Winograd F(2x2,3x3) forward and flip-mode (backward-data) vs torch,
+    including non-square channel counts (a Cw0/Cw1 stride swap in the weight
+    transform was caught exactly here)."""
+    from split_learning_amd.ops.functional import native
+    n = native()
+    torch.manual_seed(11)
+    x = torch.randn(8, ci, hw, hw, device="cuda")
+    w = torch.randn(co, ci, 3, 3, device="cuda") * 0.1
+    b = torch.randn(co, device="cuda")
+    y = n.conv2d_wino(x, w, b, 1, False)
+    ref = F.conv2d(x, w, b, 1, 1)
+    assert_close(y, ref, atol=1e-4, rtol=1e-4, what=f"wino fwd {ci}->{co}")
+
+    gy = torch.randn(8, co, hw, hw, device="cuda")
+    gx = n.conv2d_wino(gy, w, None, 1, True)
+    xr = x.clone().requires_grad_(True)
+    F.conv2d(xr, w, None, 1, 1).backward(gy)
+    assert_close(gx, xr.grad, atol=1e-4, rtol=1e-4, what=f"wino bwdd {ci}->{co}")
