@@ -48,6 +48,27 @@ static inline bool conv_pad_mode() {
   return v != 0;
 }
 
+// wino.hip
+at::Tensor conv2d_wino(const at::Tensor&, const at::Tensor&,
+                       c10::optional<at::Tensor>, int, bool);
+
+// Winograd routing (SLK_WINO=0 disables): F(2x2,3x3) beats the direct
+// implicit-GEMM kernel on the MEASURED win set only — square channel counts
+// at either end of the stack (64ch/32x32 and >=256ch tails); the 128ch
+// middle loses to V/M HBM inflation (tools/wino_check.py table in
+// profiles/SUMMARY.md).
+static inline bool wino_wins(int Ci, int Co, int H, int KH, int KW, int stride,
+                             int OH, int OW) {
+  static int v = [] {
+    const char* e = std::getenv("SLK_WINO");
+    return e ? atoi(e) : 1;
+  }();
+  if (v == 0 || KH != 3 || KW != 3 || stride != 1) return false;
+  if ((OH | OW) & 1) return false;
+  if (v >= 2) return true;  // force (sweeps)
+  return Ci == Co && (Ci >= 256 || H >= 32);
+}
+
 
 struct ConvGeom {
   int B, Ci, H, W, Co, KH, KW, OH, OW, stride, pad;
@@ -534,6 +555,12 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                       bool x_is_padded) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.scalar_type() == at::kFloat);
   TORCH_CHECK(w.is_cuda() && w.dim() == 4 && w.size(1) == x.size(1));
+  if (!x_is_padded && !conv_pad_mode()
+      && wino_wins(x.size(1), w.size(0), x.size(2), w.size(2), w.size(3),
+                   stride, x.size(2) + 2 * pad - w.size(2) + 1,
+                   x.size(3) + 2 * pad - w.size(3) + 1)) {
+    return conv2d_wino(x, w, bias, pad, /*flip=*/false);
+  }
   const bool padded = x_is_padded || conv_pad_mode();
   auto stream = c10::hip::getCurrentHIPStream().stream();
   auto wc = w.contiguous();
@@ -576,6 +603,11 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
   const int KH = w.size(2), KW = w.size(3);
   const int Ci = w.size(1), Co = w.size(0);
   const int B = gy.size(0);
+  // gx = winograd-conv of gy with rotated/transposed weights at pad' = 2-p
+  if (stride == 1 && KH == 3 && KW == 3 && pad <= 2
+      && wino_wins(Co, Ci, gy.size(2), 3, 3, 1, H, W)) {
+    return conv2d_wino(gy, w, c10::nullopt, 2 - pad, /*flip=*/true);
+  }
 
   if (conv_pad_mode() && stride == 1 && KH == KW && pad <= KH - 1) {
     // gx = valid-conv(pad(gy, K-1-p), flipT(w)): bounds-free forward gather.
