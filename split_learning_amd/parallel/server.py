@@ -44,14 +44,27 @@ def boundary_shapes(model_name: str, data_name: str, cuts: List[int],
     if key not in _SHAPE_CACHE:
         total = get_model_class(model_name, data_name).TOTAL_UNITS
         ranges = stage_ranges(cuts, n_stages, total)
-        x = _dummy_input(data_name, batch=2)
-        shapes = []
+        shapes_by_batch = []
         with torch.no_grad():
-            for (s, e) in ranges[:-1]:
-                part = build_partition(model_name, data_name, [s, e]).eval()
-                x = part(x)
-                shapes.append(tuple(x.shape[1:]))
-        _SHAPE_CACHE[key] = shapes
+            # probe TWO batch sizes: the routing contract scales the probed
+            # per-sample shape by the run batch, which silently assumes no
+            # model mixes batch into other dims — assert it instead of
+            # trusting it (round-1 VERDICT weak #7)
+            for probe_b in (2, 3):
+                x = _dummy_input(data_name, batch=probe_b)
+                shapes = []
+                for (s, e) in ranges[:-1]:
+                    part = build_partition(model_name, data_name, [s, e]).eval()
+                    x = part(x)
+                    assert x.shape[0] == probe_b, (
+                        f"{model_name} stage {s}-{e} does not keep the batch "
+                        f"axis leading: {tuple(x.shape)}")
+                    shapes.append(tuple(x.shape[1:]))
+                shapes_by_batch.append(shapes)
+        assert shapes_by_batch[0] == shapes_by_batch[1], (
+            f"{model_name}/{data_name} boundary shapes depend on batch size: "
+            f"{shapes_by_batch}")
+        _SHAPE_CACHE[key] = shapes_by_batch[0]
     return [(batch, *s) for s in _SHAPE_CACHE[key]]
 
 
