@@ -111,21 +111,24 @@ struct ConvFwdGather {
   struct KCtx {};
   __device__ KCtx prepK(int) const { return {}; }
 
-  struct ACtx { const float* row; bool valid; };
-  struct BCtx { const float* base; int ihb, iwb; bool valid; };
+  // int32 offsets against the functor's uniform (SGPR) base pointers: the
+  // pointer-context form cost ~8 VGPR of the register budget (one wave/SIMD
+  // of occupancy; see StridedGather).  Conv tensors are < 2^31 elements.
+  struct ACtx { int off; bool valid; };
+  struct BCtx { int boff, ihb, iwb; bool valid; };
 
   __device__ ACtx prepA(int, int m, bool valid, int) const {
-    return {w + (long)m * (geo.Ci * G::kh_kw(geo)), valid};
+    return {(int)((long)m * (geo.Ci * G::kh_kw(geo))), valid};
   }
   __device__ float loadA(const ACtx& c, const KCtx&, int k, bool kv) const {
-    return sel0(c.row[k], c.valid & kv);
+    return sel0(w[c.off + k], c.valid & kv);
   }
   __device__ BCtx prepB(int, int n, bool valid, int) const {
     const unsigned b = geo.d_ohow.div(n);
     const unsigned rem = geo.d_ohow.mod(n, b);
     const unsigned oh = geo.d_ow.div(rem);
     const unsigned ow = geo.d_ow.mod(rem, oh);
-    return {x + (long)b * geo.Ci * geo.H * geo.W,
+    return {(int)((long)b * geo.Ci * geo.H * geo.W),
             (int)oh * G::stride(geo) - geo.pad,
             (int)ow * G::stride(geo) - geo.pad, valid};
   }
@@ -137,7 +140,7 @@ struct ConvFwdGather {
     const bool in = (unsigned)ih < (unsigned)geo.H && (unsigned)iw < (unsigned)geo.W;
     const int ihc = in ? ih : 0;
     const int iwc = in ? iw : 0;
-    const float v = c.base[((long)ci * geo.H + ihc) * geo.W + iwc];
+    const float v = x[c.boff + (ci * geo.H + ihc) * geo.W + iwc];
     return sel0(v, c.valid & kv & in);
   }
 };
@@ -210,7 +213,7 @@ struct ConvBwdDataGather {
   __device__ KCtx prepK(int) const { return {}; }
 
   struct ACtx { int m; bool valid; };
-  struct BCtx { const float* base; int ihp, iwp; bool valid; };
+  struct BCtx { int boff, ihp, iwp; bool valid; };
 
   __device__ ACtx prepA(int, int m, bool valid, int) const { return {m, valid}; }
   __device__ float loadA(const ACtx& c, const KCtx&, int k, bool kv) const {
@@ -225,7 +228,7 @@ struct ConvBwdDataGather {
     const unsigned rem = geo.d_hw.mod(n, b);
     const unsigned ih = geo.d_w.div(rem);
     const unsigned iw = geo.d_w.mod(rem, ih);
-    return {gy + (long)b * geo.Co * geo.OH * geo.OW,
+    return {(int)((long)b * geo.Co * geo.OH * geo.OW),
             (int)ih + geo.pad, (int)iw + geo.pad, valid};
   }
   __device__ float loadB(const BCtx& c, const KCtx&, int k, bool kv) const {
@@ -256,7 +259,7 @@ struct ConvBwdDataGather {
     ok = ok && (unsigned)oh < (unsigned)geo.OH && (unsigned)ow < (unsigned)geo.OW;
     const int ohc = ok ? oh : 0;
     const int owc = ok ? ow : 0;
-    const float v = c.base[((long)co * geo.OH + ohc) * geo.OW + owc];
+    const float v = gy[c.boff + (co * geo.OH + ohc) * geo.OW + owc];
     return sel0(v, c.valid & kv & ok);
   }
 };
@@ -290,22 +293,22 @@ struct ConvBwdWeightGather {
   ConvGeom geo;
   bool fast;        // OH*OW % 16 == 0: window-hoisted A addressing
 
-  struct KCtx { long a_off; };
+  struct KCtx { int a_off; };
   __device__ KCtx prepK(int k0) const {
     if (!fast) return {0};
     const unsigned b0 = geo.d_ohow.div(k0);
     const unsigned rem0 = geo.d_ohow.mod(k0, b0);
-    return {(long)b0 * geo.Co * (geo.OH * geo.OW) + rem0};
+    return {(int)((long)b0 * geo.Co * (geo.OH * geo.OW) + rem0)};
   }
 
-  struct ACtx { const float* rowdk; int m; bool valid; };
+  struct ACtx { int rowdk; int m; bool valid; };
   __device__ ACtx prepA(int, int m, bool valid, int dk) const {
-    return {gy + (long)m * (geo.OH * geo.OW) + dk, m, valid};
+    return {(int)((long)m * (geo.OH * geo.OW) + dk), m, valid};
   }
   __device__ float loadA(const ACtx& c, const KCtx& kc, int k, bool kv) const {
     float v;
     if (fast) {
-      v = c.rowdk[kc.a_off];     // per-lane hoisted row + per-tile scalar
+      v = gy[c.rowdk + kc.a_off];  // per-lane hoisted row + per-tile scalar
     } else {
       const unsigned b = geo.d_ohow.div(k);
       const unsigned rem = geo.d_ohow.mod(k, b);
@@ -314,11 +317,11 @@ struct ConvBwdWeightGather {
     return sel0(v, c.valid & kv);
   }
 
-  struct BCtx { long ciHW; int kh, kw; bool valid; };
+  struct BCtx { int ciHW; short kh, kw; bool valid; };
   __device__ BCtx prepB(int, int n, bool valid, int) const {
     int ci, kh, kw;
     G::dk(geo, n, ci, kh, kw);
-    return {(long)ci * geo.H * geo.W, kh, kw, valid};
+    return {ci * geo.H * geo.W, (short)kh, (short)kw, valid};
   }
   __device__ float loadB(const BCtx& c, const KCtx&, int k, bool kv) const {
     const unsigned b = geo.d_ohow.div(k);
@@ -336,7 +339,7 @@ struct ConvBwdWeightGather {
     const int ihc = in ? ih : 0;
     const int iwc = in ? iw : 0;
     const float v = x[(long)b * geo.Ci * geo.H * geo.W + c.ciHW
-                      + (long)ihc * geo.W + iwc];
+                      + ihc * geo.W + iwc];
     return sel0(v, c.valid & kv & in);
   }
 };
