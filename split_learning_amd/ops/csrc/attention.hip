@@ -39,8 +39,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     float* __restrict__ probs, unsigned char* __restrict__ mask, int S, int hd,
     float scale, float p, float inv_keep, uint64_t seed,
     const long* __restrict__ offset_ptr) {
-  // device-side philox offset (graph-safe: advances under hipGraph replay)
+  // device-side philox offset (graph-safe: advances under hipGraph replay),
+  // folded with the seed into one 32-bit stream id for the cheap mask hash
   const uint64_t offset = offset_ptr ? (uint64_t)offset_ptr[0] : 0;
+  const uint32_t rng_stream = (uint32_t)(slk_mix64(seed ^ slk_mix64(offset)) >> 32);
   extern __shared__ float lds[];
   float* ldsK = lds;                      // [S][hd+1]
   float* ldsV = ldsK + S * (hd + 1);      // [S][hd+1]
@@ -89,7 +91,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       const float pv = e0 * inv;
       float pd = pv;
       if (DROPOUT) {
-        const bool keep = slk_uniform(seed, offset, ridx + c0) >= p;
+        const bool keep = slk_uniform32(rng_stream, (uint32_t)(ridx + c0)) >= p;
         pd = keep ? pv * inv_keep : 0.f;
         mask[ridx + c0] = keep;
       }
@@ -100,7 +102,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       const float pv = e1 * inv;
       float pd = pv;
       if (DROPOUT) {
-        const bool keep = slk_uniform(seed, offset, ridx + c1) >= p;
+        const bool keep = slk_uniform32(rng_stream, (uint32_t)(ridx + c1)) >= p;
         pd = keep ? pv * inv_keep : 0.f;
         mask[ridx + c1] = keep;
       }

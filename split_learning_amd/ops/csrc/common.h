@@ -59,6 +59,21 @@ __device__ __forceinline__ float slk_uniform(uint64_t seed, uint64_t offset,
   return (float)(h >> 40) * (1.0f / 16777216.0f);
 }
 
+// 32-bit mask hash (lowbias32) for dropout draws INSIDE compute kernels:
+// ~6 VALU vs ~25 for the 64-bit path (two splitmix64 = four 64-bit
+// multiplies) — the difference made the fused attention kernel VALU-bound
+// at BERT scale.  Fold (seed, offset) into one 32-bit stream id host/SALU-
+// side via slk_mix64 first.
+__device__ __forceinline__ float slk_uniform32(uint32_t stream, uint32_t idx) {
+  uint32_t x = idx * 0x9E3779B9u + stream;
+  x ^= x >> 16;
+  x *= 0x7FEB352Du;
+  x ^= x >> 15;
+  x *= 0x846CA68Bu;
+  x ^= x >> 16;
+  return (float)(x >> 8) * (1.0f / 16777216.0f);
+}
+
 // Fast integer division by a runtime constant (one 64-bit mul + shift).
 // Valid for dividend < 2^24 and divisor < 2^16 (all tensor-index math here):
 // m = floor(2^40/d)+1; q = (n*m) >> 40 == n/d exactly when n*d < 2^40.

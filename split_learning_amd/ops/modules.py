@@ -174,15 +174,19 @@ class HipMultiheadAttention(nn.MultiheadAttention):
         return out, None
 
 
+_FATTN_FORCE = os.environ.get("SLK_FATTN", "0") == "1"
+
+
 def _fused_attn_worth_it(bh, s_len, dropout_p, training) -> bool:
-    """The fused kernel wins when launches dominate; with in-kernel dropout
-    its per-element RNG (two splitmix64 hashes) makes it VALU-bound, and at
-    BERT scale (BH=384, S=128 -> 6.3M mask draws) the routed GEMM path
-    measured ~10x faster (profiles/SUMMARY.md round 2).  Dropout-off: always
-    fused.  Dropout-on: fused only for small score matrices."""
-    if dropout_p == 0.0 or not training:
-        return True
-    return bh * s_len * s_len <= (1 << 21)
+    """Round-2 A/B (profiles/SUMMARY.md): the one-launch fused SDPA kernel
+    LOSES to the routed MFMA-GEMM + fused-softmax(+dropout) chain at every
+    model shape, dropout or not — its grid is BH workgroups (<= 384 on the
+    zoo) with scalar-FMA row sweeps, so the chip runs underfilled:
+    BERT 308 vs 63 us, KWT 200 vs 56, ViT 78 vs 55.  The GEMM chain is also
+    all hand-written kernels, so routing there keeps the native path.  The
+    fused kernel stays available (SLK_FATTN=1 + kernel tests) pending the
+    roadmap rewrite (block-per-(bh, row-tile) with MFMA scores/PV)."""
+    return _FATTN_FORCE
 
 
 def attention_core(q, k, v, dropout_p: float = 0.0, training: bool = False,
