@@ -51,12 +51,26 @@ static inline bool conv_pad_mode() {
 // wino.hip
 at::Tensor conv2d_wino(const at::Tensor&, const at::Tensor&,
                        c10::optional<at::Tensor>, int, bool);
+at::Tensor conv2d_wino_bwdw(const at::Tensor&, const at::Tensor&, int);
 
 // Winograd routing (SLK_WINO=0 disables): F(2x2,3x3) beats the direct
 // implicit-GEMM kernel on the MEASURED win set only — square channel counts
 // at either end of the stack (64ch/32x32 and >=256ch tails); the 128ch
 // middle loses to V/M HBM inflation (tools/wino_check.py table in
 // profiles/SUMMARY.md).
+static inline bool wino_wins_bwdw(int Ci, int Co, int H, int OH, int OW) {
+  static int v = [] {
+    const char* e = std::getenv("SLK_WINO");
+    return e ? atoi(e) : 1;
+  }();
+  if (v == 0) return false;
+  if ((OH | OW) & 1) return false;
+  if (v >= 2) return true;
+  // measured win set (tools/wino_check.py --bwdw): TO CALIBRATE — start
+  // with every eligible 3x3 s1 shape and trim by measurement
+  return true;
+}
+
 static inline bool wino_wins(int Ci, int Co, int H, int KH, int KW, int stride,
                              int OH, int OW) {
   static int v = [] {
@@ -658,6 +672,17 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
 at::Tensor conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x, int KH,
                              int KW, int stride, int pad, bool x_is_padded) {
   TORCH_CHECK(gy.is_cuda() && x.is_cuda() && gy.scalar_type() == at::kFloat);
+  // Winograd weight-gradient: dU[f] = (A gy A^T)[f] @ V_x[f]^T then
+  // gw = G^T dU G — replaces the direct gather's pathological tall-K GEMM
+  // ([Co, Ci*9, B*OH*OW]: 9 n-tiles at conv1_2) with 16 well-shaped
+  // frequency GEMMs.  Same routing guard as fwd (bwd-weight win set is
+  // broader; SLK_WINO applies).
+  if (!x_is_padded && !conv_pad_mode() && stride == 1 && KH == 3 && KW == 3
+      && pad <= 2
+      && wino_wins_bwdw(x.size(1), gy.size(1), x.size(2), gy.size(2),
+                        gy.size(3))) {
+    return conv2d_wino_bwdw(gy, x, pad);
+  }
   const bool padded = x_is_padded || conv_pad_mode();
   auto gyc = gy.contiguous();
   auto stream = c10::hip::getCurrentHIPStream().stream();

@@ -160,6 +160,74 @@ __global__ void wino_out_kernel(const float* __restrict__ Mm,
   }
 }
 
+// ---- backward-weight --------------------------------------------------
+// dY arrives per 2x2 output tile; dM = A dY A^T lifts it to the 4x4
+// frequency domain (A = [[1,0],[1,1],[1,-1],[0,-1]]), then
+// dU[f][Co][Ci] = dM[f][Co,T] @ V_x[f][Ci,T]^T (16 frequency GEMMs over the
+// tile axis — a far better GEMM shape than the direct gather's
+// [Co, Ci*9, B*OH*OW] tall-K formulation), and gw = G^T dU G.
+__global__ void wino_gyA_kernel(const float* __restrict__ gy,
+                                float* __restrict__ Wg, int B, int Co, int OH,
+                                int OW, int tH, int tW, FastDiv d_T,
+                                FastDiv d_thw, FastDiv d_tw) {
+  const int T = B * tH * tW;
+  const long total = (long)Co * T;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const unsigned co = d_T.div((unsigned)i);
+    const unsigned t = d_T.mod((unsigned)i, co);
+    const unsigned b = d_thw.div(t);
+    const unsigned rem = d_thw.mod(t, b);
+    const unsigned th = d_tw.div(rem);
+    const unsigned tw = d_tw.mod(rem, th);
+    const float* gp = gy + ((long)b * Co + co) * OH * OW
+                      + (long)th * 2 * OW + tw * 2;
+    const float g00 = gp[0], g01 = gp[1], g10 = gp[OW], g11 = gp[OW + 1];
+    // rows of A gy (4x2): [g0j], [g0j+g1j], [g0j-g1j], [-g1j]
+    float r0c0 = g00, r0c1 = g01;
+    float r1c0 = g00 + g10, r1c1 = g01 + g11;
+    float r2c0 = g00 - g10, r2c1 = g01 - g11;
+    float r3c0 = -g10, r3c1 = -g11;
+    // columns: dM[a][.] = [c0], [c0+c1], [c0-c1], [-c1]
+    #pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      const float c0 = a == 0 ? r0c0 : a == 1 ? r1c0 : a == 2 ? r2c0 : r3c0;
+      const float c1 = a == 0 ? r0c1 : a == 1 ? r1c1 : a == 2 ? r2c1 : r3c1;
+      Wg[((long)(a * 4 + 0) * Co + co) * T + t] = c0;
+      Wg[((long)(a * 4 + 1) * Co + co) * T + t] = c0 + c1;
+      Wg[((long)(a * 4 + 2) * Co + co) * T + t] = c0 - c1;
+      Wg[((long)(a * 4 + 3) * Co + co) * T + t] = -c1;
+    }
+  }
+}
+
+__global__ void wino_gw_kernel(const float* __restrict__ dU,
+                               float* __restrict__ gw, int Co, int Ci) {
+  const long total = (long)Co * Ci;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    float u[4][4];
+    #pragma unroll
+    for (int f = 0; f < 16; ++f) u[f >> 2][f & 3] = dU[(long)f * total + i];
+    // t = G^T u (3x4), G = [[1,0,0],[.5,.5,.5],[.5,-.5,.5],[0,0,1]]
+    float t[3][4];
+    #pragma unroll
+    for (int b = 0; b < 4; ++b) {
+      t[0][b] = u[0][b] + 0.5f * (u[1][b] + u[2][b]);
+      t[1][b] = 0.5f * (u[1][b] - u[2][b]);
+      t[2][b] = 0.5f * (u[1][b] + u[2][b]) + u[3][b];
+    }
+    #pragma unroll
+    for (int r = 0; r < 3; ++r) {
+      gw[i * 9 + r * 3 + 0] = t[r][0] + 0.5f * (t[r][1] + t[r][2]);
+      gw[i * 9 + r * 3 + 1] = 0.5f * (t[r][1] - t[r][2]);
+      gw[i * 9 + r * 3 + 2] = 0.5f * (t[r][1] + t[r][2]) + t[r][3];
+    }
+  }
+}
+
 // host entry: 3x3 stride-1 conv via F(2x2,3x3); flip=true computes the
 // bwd-data conv (weights rotated + Co/Ci transposed, pad' = 2 - pad)
 at::Tensor matmul_f32(const at::Tensor&, const at::Tensor&, bool, bool,
@@ -224,6 +292,52 @@ at::Tensor conv2d_wino(const at::Tensor& x, const at::Tensor& w,
                        B, Co, OH, OW, tH, tW, d_T, d_thw, d_tw);
   }
   return y;
+}
+
+at::Tensor conv2d_wino_bwdw(const at::Tensor& gy, const at::Tensor& x,
+                            int pad) {
+  TORCH_CHECK(gy.is_cuda() && x.is_cuda() && gy.dim() == 4 && x.dim() == 4);
+  auto gyc = gy.contiguous();
+  auto xc = x.contiguous();
+  const int B = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int Co = gy.size(1), OH = gy.size(2), OW = gy.size(3);
+  TORCH_CHECK(OH == H + 2 * pad - 2 && OW == W + 2 * pad - 2,
+              "wino_bwdw geometry");
+  TORCH_CHECK(OH % 2 == 0 && OW % 2 == 0, "wino_bwdw: even output dims only");
+  const int tH = OH / 2, tW = OW / 2;
+  const int T = B * tH * tW;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  FastDiv d_T, d_thw, d_tw;
+  d_T.init(T);
+  d_thw.init(tH * tW);
+  d_tw.init(tW);
+
+  auto Wg = at::empty({16, Co, T}, x.options());
+  auto Vx = at::empty({16, Ci, T}, x.options());
+  {
+    const long tot = (long)Co * T;
+    const int grid = (int)std::min<long>((tot + 255) / 256, 8192);
+    hipLaunchKernelGGL(wino_gyA_kernel, dim3(grid), dim3(256), 0, stream,
+                       gyc.data_ptr<float>(), Wg.data_ptr<float>(), B, Co, OH,
+                       OW, tH, tW, d_T, d_thw, d_tw);
+  }
+  {
+    const long tot = (long)Ci * T;
+    const int grid = (int)std::min<long>((tot + 255) / 256, 8192);
+    hipLaunchKernelGGL(wino_in_kernel, dim3(grid), dim3(256), 0, stream,
+                       xc.data_ptr<float>(), Vx.data_ptr<float>(), B, Ci, H, W,
+                       tH, tW, pad, d_T, d_thw, d_tw);
+  }
+  // dU[f][Co][Ci] = Wg[f] @ Vx[f]^T : 16 frequency GEMMs, K = tiles
+  auto dU = matmul_f32(Wg, Vx, false, true, c10::nullopt, false);
+  auto gw = at::empty({Co, Ci, 3, 3}, x.options());
+  {
+    const long tot = (long)Co * Ci;
+    const int grid = (int)std::min<long>((tot + 255) / 256, 4096);
+    hipLaunchKernelGGL(wino_gw_kernel, dim3(grid), dim3(256), 0, stream,
+                       dU.data_ptr<float>(), gw.data_ptr<float>(), Co, Ci);
+  }
+  return gw;
 }
 
 }  // namespace slk
