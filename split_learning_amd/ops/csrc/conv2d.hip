@@ -66,9 +66,10 @@ static inline bool wino_wins_bwdw(int Ci, int Co, int H, int OH, int OW) {
   if (v == 0) return false;
   if ((OH | OW) & 1) return false;
   if (v >= 2) return true;
-  // measured win set (tools/wino_check.py --bwdw): TO CALIBRATE — start
-  // with every eligible 3x3 s1 shape and trim by measurement
-  return true;
+  // measured win set (tools/wino_check.py under SLK_WINO=0): square channel
+  // counts win everywhere EXCEPT the 4x4 layers, where T = B*4 makes the
+  // frequency GEMMs too small (512ch 4x4: 62 vs 48 us direct)
+  return Ci == Co && H != 4;
 }
 
 static inline bool wino_wins(int Ci, int Co, int H, int KH, int KW, int stride,
