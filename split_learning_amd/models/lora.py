@@ -41,7 +41,8 @@ class LoRALinear(nn.Module):
         else:
             xa = xd @ self.lora_A.t()
             delta = xa @ self.lora_B.t()
-        return y + delta * self.scaling
+        # fused scale+add (one aten kernel instead of mul + add)
+        return torch.add(y, delta, alpha=self.scaling)
 
     def merged_weight(self) -> torch.Tensor:
         return self.base.weight + (self.lora_B @ self.lora_A) * self.scaling
