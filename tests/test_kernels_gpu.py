@@ -589,3 +589,37 @@ def test_conv2d_winograd(ci, hw, co):
     xr = x.clone().requires_grad_(True)
     F.conv2d(xr, w, None, 1, 1).backward(gy)
     assert_close(gx, xr.grad, atol=1e-4, rtol=1e-4, what=f"wino bwdd {ci}->{co}")
+
+
+@pytest.mark.gpu
+def test_conv_bias_grad_zero_under_bn():
+    """A Conv2d feeding a training-mode BatchNorm gets an analytically-zero
+    bias gradient (sum(xhat) == 0); the fused module path skips the reduction
+    and must agree with torch's computed (noise-level) gradient."""
+    import torch.nn as tnn
+    from split_learning_amd.models.partitioned import SequentialUnits
+
+    class Tiny(SequentialUnits):
+        TOTAL_UNITS = 3
+
+        @classmethod
+        def unit_factories(cls):
+            from split_learning_amd.ops.modules import (HipBatchNorm2d,
+                                                        HipConv2d, HipReLU)
+            return {1: lambda: HipConv2d(8, 16, 3, padding=1),
+                    2: lambda: HipBatchNorm2d(16),
+                    3: lambda: HipReLU()}
+
+    m = Tiny(0, 3).cuda().train()
+    x = torch.randn(16, 8, 10, 10, device="cuda")
+    m(x).sum().backward()
+    gb = m.layer1.bias.grad
+    assert gb is not None
+    assert float(gb.abs().max()) == 0.0  # fused path: exact zeros
+
+    # torch reference computes the same thing as numerical noise
+    ref = tnn.Sequential(tnn.Conv2d(8, 16, 3, padding=1), tnn.BatchNorm2d(16),
+                         tnn.ReLU()).cuda().train()
+    xr = x.clone()
+    ref(xr).sum().backward()
+    assert float(ref[0].bias.grad.abs().max()) < 1e-3
