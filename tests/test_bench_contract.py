@@ -64,3 +64,18 @@ def test_bench_n2_torchrun_cpu_contract():
         capture_output=True, text=True, cwd=REPO, timeout=850)
     assert p.returncode == 0, p.stderr[-2000:]
     _check(_json_line(p.stdout), 2)
+
+
+@pytest.mark.timeout(900)
+def test_bench_n4_torchrun_cpu_contract():
+    """4-rank rendezvous shape (2 pipelines) on gloo: the same launch the
+    driver uses for the SCALE curve, so a multi-pipeline regression (rank
+    mapping, eager comm warmup ordering, poison-drain) is caught on CPU."""
+    p = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29423", "bench.py", "--gpus", "4",
+         "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, cwd=REPO, timeout=850)
+    assert p.returncode == 0, p.stderr[-2000:]
+    _check(_json_line(p.stdout), 4)

@@ -265,3 +265,35 @@ def test_first_stage_stall_raises(tmp_path):
     with pytest.raises(StageStallError) as ei:
         train_first_stage(ctx)
     assert "first" in str(ei.value)
+
+
+def _bcast_unit_worker(rank, world, port, tmpdir):
+    import torch.distributed as dist
+    from split_learning_amd.models import build_partition
+    from split_learning_amd.parallel import bcast
+    dist.init_process_group("gloo", init_method=f"tcp://127.0.0.1:{port}",
+                            rank=rank, world_size=world)
+    full = build_partition("VGG16", "CIFAR10", [0, 0]).state_dict() \
+        if rank == 0 else None
+    out = bcast.broadcast_full_state("VGG16", "CIFAR10", full,
+                                     torch.device("cpu"))
+    ref = build_partition("VGG16", "CIFAR10", [0, 0]).state_dict()
+    assert set(out.keys()) == set(ref.keys())
+    for k, v in out.items():
+        assert v.shape == ref[k].shape and v.dtype == ref[k].dtype, k
+    # round-trip exactness on rank 0 (fp32 path is bit-exact; int buffers ride
+    # as fp32, exact below 2^24)
+    if rank == 0:
+        for k in full:
+            assert torch.equal(out[k], full[k]), k
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_param_broadcast_roundtrip(tmp_path):
+    """bcast.broadcast_full_state: every rank reconstructs the exact full
+    state dict (keys, shapes, dtypes; bit-exact values on the source)."""
+    port = _free_port()
+    mp.spawn(_bcast_unit_worker, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
