@@ -11,6 +11,8 @@ at::Tensor colsum_f32(const at::Tensor&);
 at::Tensor conv2d_wino(const at::Tensor&, const at::Tensor&,
                        c10::optional<at::Tensor>, int, bool);
 at::Tensor conv2d_wino_bwdw(const at::Tensor&, const at::Tensor&, int);
+at::Tensor conv2d_wino_fused(const at::Tensor&, const at::Tensor&,
+                             c10::optional<at::Tensor>, int, bool);
 // conv2d.hip
 at::Tensor pad_nchw(const at::Tensor&, int);
 at::Tensor conv2d_fwd(const at::Tensor&, const at::Tensor&, c10::optional<at::Tensor>,
@@ -91,6 +93,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("flip") = false);
   m.def("conv2d_wino_bwdw", &slk::conv2d_wino_bwdw, py::arg("gy"),
         py::arg("x"), py::arg("pad") = 1);
+  m.def("conv2d_wino_fused", &slk::conv2d_wino_fused, py::arg("x"),
+        py::arg("w"), py::arg("bias") = py::none(), py::arg("pad") = 1,
+        py::arg("flip") = false);
   m.def("pad_nchw", &slk::pad_nchw);
   m.def("conv2d_fwd", &slk::conv2d_fwd, py::arg("x"), py::arg("w"),
         py::arg("bias"), py::arg("stride"), py::arg("pad"),

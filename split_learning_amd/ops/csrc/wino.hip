@@ -294,6 +294,160 @@ at::Tensor conv2d_wino(const at::Tensor& x, const at::Tensor& w,
   return y;
 }
 
+
+// ---- fused F(2x2,3x3) kernel ------------------------------------------
+// One kernel: input transform in-register, 16-frequency MFMA accumulation,
+// output transform in the epilogue — no V/M HBM round trip (the unfused
+// pipeline's 4x data inflation).  A block owns a (32 co x 32 tile) output
+// patch across ALL 16 frequencies, so each 4x4 input patch is read once and
+// feeds every frequency.  Per wave: 16x16 (co, t) fragment x 16 freq = 16
+// f32x4 accumulators (64 AGPR).  K-loop over ci in steps of 8.
+// Weights arrive PRE-TRANSFORMED (U[16][Co][Ci], wino_wt_kernel — tiny).
+constexpr int WF_CO = 32;   // block co tile
+constexpr int WF_T = 32;    // block tile-column tile
+constexpr int WF_CK = 8;    // ci step
+// LDS row strides padded to dodge the worst bank conflicts
+constexpr int WF_ULD = WF_CO + 1;   // U_lds[f][ci][co]
+constexpr int WF_VLD = WF_T + 1;    // V_lds[f][ci][t]
+
+__global__ __launch_bounds__(256) void wino_fused_kernel(
+    const float* __restrict__ x, const float* __restrict__ U,
+    const float* __restrict__ bias, float* __restrict__ y, int B, int Ci,
+    int H, int W, int Co, int OH, int OW, int tH, int tW, int pad,
+    FastDiv d_thw, FastDiv d_tw) {
+  // single-buffer two-barrier loop: double buffering would cost 67.6 KB of
+  // LDS (2 blocks/CU); at 33.8 KB four blocks fit and the long 32-MFMA
+  // phase gives co-resident waves the latency cover instead
+  __shared__ float Ul[16 * WF_CK * WF_ULD];
+  __shared__ float Vl[16 * WF_CK * WF_VLD];
+
+  const int T = B * tH * tW;
+  const int co0 = blockIdx.y * WF_CO;
+  const int t0 = blockIdx.x * WF_T;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wco = (wid >> 1) * 16;   // wave offset in the 32-co tile
+  const int wt = (wid & 1) * 16;     // wave offset in the 32-t tile
+  const int frag_r = lane >> 4;      // k sub-index (0..3)
+  const int frag_c = lane & 15;
+
+  f32x4 acc[16];
+  #pragma unroll
+  for (int f = 0; f < 16; ++f) acc[f] = f32x4{};
+
+  // staging assignments -----------------------------------------------
+  // V: thread -> one (ci, t) pair: ci = tid >> 5 (8), t = tid & 31 (32)
+  const int v_ci = tid >> 5;
+  const int v_t = tid & 31;
+  const int tg = t0 + v_t;
+  const unsigned vb = d_thw.div((unsigned)tg);
+  const unsigned vrem = d_thw.mod((unsigned)tg, vb);
+  const unsigned vth = d_tw.div(vrem);
+  const unsigned vtw = d_tw.mod(vrem, vth);
+  const int ih0 = (int)vth * 2 - pad;
+  const int iw0 = (int)vtw * 2 - pad;
+  const long xplane = ((long)vb * Ci + v_ci) * H * W;  // + ciстеп*H*W later
+  // U: each thread stages 16 elements (256 threads x 16 = 16f x 32co x 8ci)
+  const int u_lin = tid;
+
+  auto load_stage = [&](int ci0) {
+    // ---- V: 4x4 patch -> B^T d B, 16 freq values
+    {
+      const float* xp = x + xplane + (long)ci0 * H * W;
+      float d[4][4];
+      #pragma unroll
+      for (int a = 0; a < 4; ++a) {
+        const int ih = ih0 + a;
+        const bool hv = (unsigned)ih < (unsigned)H;
+        #pragma unroll
+        for (int bb = 0; bb < 4; ++bb) {
+          const int iw = iw0 + bb;
+          const bool v = hv && (unsigned)iw < (unsigned)W;
+          d[a][bb] = v ? xp[(long)ih * W + iw] : 0.f;
+        }
+      }
+      float u[4][4];
+      #pragma unroll
+      for (int bb = 0; bb < 4; ++bb) {
+        u[0][bb] = d[0][bb] - d[2][bb];
+        u[1][bb] = d[1][bb] + d[2][bb];
+        u[2][bb] = d[2][bb] - d[1][bb];
+        u[3][bb] = d[1][bb] - d[3][bb];
+      }
+      #pragma unroll
+      for (int a = 0; a < 4; ++a) {
+        const float v0 = u[a][0] - u[a][2];
+        const float v1 = u[a][1] + u[a][2];
+        const float v2 = u[a][2] - u[a][1];
+        const float v3 = u[a][1] - u[a][3];
+        Vl[((a * 4 + 0) * WF_CK + v_ci) * WF_VLD + v_t] = v0;
+        Vl[((a * 4 + 1) * WF_CK + v_ci) * WF_VLD + v_t] = v1;
+        Vl[((a * 4 + 2) * WF_CK + v_ci) * WF_VLD + v_t] = v2;
+        Vl[((a * 4 + 3) * WF_CK + v_ci) * WF_VLD + v_t] = v3;
+      }
+    }
+    // ---- U: 16 x 32 x 8 staged, 16 elements per thread
+    {
+      #pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const int e = u_lin + j * 256;        // 0..4095
+        const int f = e >> 8;                 // 0..15
+        const int r = e & 255;                // co*8 + ci
+        const int co = r >> 3;
+        const int ci = r & 7;
+        Ul[(f * WF_CK + ci) * WF_ULD + co] =
+            U[((long)f * Co + co0 + co) * Ci + ci0 + ci];
+      }
+    }
+  };
+
+  for (int ci0 = 0; ci0 < Ci; ci0 += WF_CK) {
+    load_stage(ci0);
+    __syncthreads();
+    #pragma unroll
+    for (int f = 0; f < 16; ++f) {
+      #pragma unroll
+      for (int kk = 0; kk < WF_CK / 4; ++kk) {
+        const int kr = kk * 4 + frag_r;
+        const float a = Ul[(f * WF_CK + kr) * WF_ULD + wco + frag_c];
+        const float b = Vl[(f * WF_CK + kr) * WF_VLD + wt + frag_c];
+        acc[f] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[f], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: lane holds, per freq, rows co = wco + frag_r*4 + i (i<4) at
+  // col t = wt + frag_c -> all 16 freqs of each (co, t): y = A^T M A + bias
+  #pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int co = co0 + wco + frag_r * 4 + i;
+    const int t = t0 + wt + frag_c;
+    if (co >= Co || t >= T) continue;
+    float m[4][4];
+    #pragma unroll
+    for (int f = 0; f < 16; ++f) m[f >> 2][f & 3] = acc[f][i];
+    float u0[4], u1[4];
+    #pragma unroll
+    for (int bb = 0; bb < 4; ++bb) {
+      u0[bb] = m[0][bb] + m[1][bb] + m[2][bb];
+      u1[bb] = m[1][bb] - m[2][bb] - m[3][bb];
+    }
+    const float bv = bias != nullptr ? bias[co] : 0.f;
+    const unsigned b = d_thw.div((unsigned)t);
+    const unsigned rem = d_thw.mod((unsigned)t, b);
+    const unsigned th = d_tw.div(rem);
+    const unsigned tw = d_tw.mod(rem, th);
+    float* yp = y + ((long)b * Co + co) * OH * OW + (long)th * 2 * OW + tw * 2;
+    yp[0] = u0[0] + u0[1] + u0[2] + bv;
+    yp[1] = u0[1] - u0[2] - u0[3] + bv;
+    yp[OW] = u1[0] + u1[1] + u1[2] + bv;
+    yp[OW + 1] = u1[1] - u1[2] - u1[3] + bv;
+  }
+}
+
 at::Tensor conv2d_wino_bwdw(const at::Tensor& gy, const at::Tensor& x,
                             int pad) {
   TORCH_CHECK(gy.is_cuda() && x.is_cuda() && gy.dim() == 4 && x.dim() == 4);
@@ -338,6 +492,54 @@ at::Tensor conv2d_wino_bwdw(const at::Tensor& gy, const at::Tensor& x,
                        dU.data_ptr<float>(), gw.data_ptr<float>(), Co, Ci);
   }
   return gw;
+}
+
+// fully-fused variant: transforms in-kernel, no V/M round trip.  Shape
+// requirements: Co % 32 == 0, T % 32 == 0, Ci % 8 == 0.  SLK_WINO_FUSED=0
+// falls back to the unfused pipeline (A/B).
+at::Tensor conv2d_wino_fused(const at::Tensor& x, const at::Tensor& w,
+                             c10::optional<at::Tensor> bias, int pad,
+                             bool flip) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.scalar_type() == at::kFloat);
+  TORCH_CHECK(w.size(2) == 3 && w.size(3) == 3, "wino_fused: 3x3 only");
+  auto xc = x.contiguous();
+  auto wc = w.contiguous();
+  const int B = x.size(0), H = x.size(2), W = x.size(3);
+  const int Co = flip ? w.size(1) : w.size(0);
+  const int Ci = flip ? w.size(0) : w.size(1);
+  TORCH_CHECK(x.size(1) == Ci, "wino_fused: channel mismatch");
+  const int OH = H + 2 * pad - 2, OW = W + 2 * pad - 2;
+  const int tH = OH / 2, tW = OW / 2;
+  const int T = B * tH * tW;
+  TORCH_CHECK(Co % 32 == 0 && T % 32 == 0 && Ci % 8 == 0 && OH % 2 == 0
+              && OW % 2 == 0, "wino_fused shape requirements");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+
+  auto U = at::empty({16, Co, Ci}, x.options());
+  {
+    const long tot = (long)Co * Ci;
+    const int grid = (int)std::min<long>((tot + 255) / 256, 4096);
+    if (flip) {
+      hipLaunchKernelGGL(wino_wt_kernel<true>, dim3(grid), dim3(256), 0,
+                         stream, wc.data_ptr<float>(), U.data_ptr<float>(),
+                         (int)w.size(0), (int)w.size(1));
+    } else {
+      hipLaunchKernelGGL(wino_wt_kernel<false>, dim3(grid), dim3(256), 0,
+                         stream, wc.data_ptr<float>(), U.data_ptr<float>(),
+                         (int)w.size(0), (int)w.size(1));
+    }
+  }
+  auto y = at::empty({B, Co, OH, OW}, x.options());
+  FastDiv d_thw, d_tw;
+  d_thw.init(tH * tW);
+  d_tw.init(tW);
+  dim3 grid(T / 32, Co / 32);
+  hipLaunchKernelGGL(wino_fused_kernel, grid, dim3(256), 0, stream,
+                     xc.data_ptr<float>(), U.data_ptr<float>(),
+                     bias.has_value() ? bias->data_ptr<float>() : nullptr,
+                     y.data_ptr<float>(), B, Ci, H, W, Co, OH, OW, tH, tW,
+                     pad, d_thw, d_tw);
+  return y;
 }
 
 }  // namespace slk
