@@ -310,7 +310,7 @@ constexpr int WF_CK = 8;    // ci step
 constexpr int WF_ULD = WF_CO + 1;   // U_lds[f][ci][co]
 constexpr int WF_VLD = WF_T + 1;    // V_lds[f][ci][t]
 
-__global__ __launch_bounds__(256) void wino_fused_kernel(
+__global__ __launch_bounds__(256, 2) void wino_fused_kernel(
     const float* __restrict__ x, const float* __restrict__ U,
     const float* __restrict__ bias, float* __restrict__ y, int B, int Ci,
     int H, int W, int Co, int OH, int OW, int tH, int tW, int pad,
@@ -352,60 +352,64 @@ __global__ __launch_bounds__(256) void wino_fused_kernel(
   // U: each thread stages 16 elements (256 threads x 16 = 16f x 32co x 8ci)
   const int u_lin = tid;
 
-  auto load_stage = [&](int ci0) {
-    // ---- V: 4x4 patch -> B^T d B, 16 freq values
-    {
-      const float* xp = x + xplane + (long)ci0 * H * W;
-      float d[4][4];
-      #pragma unroll
-      for (int a = 0; a < 4; ++a) {
-        const int ih = ih0 + a;
-        const bool hv = (unsigned)ih < (unsigned)H;
-        #pragma unroll
-        for (int bb = 0; bb < 4; ++bb) {
-          const int iw = iw0 + bb;
-          const bool v = hv && (unsigned)iw < (unsigned)W;
-          d[a][bb] = v ? xp[(long)ih * W + iw] : 0.f;
-        }
-      }
-      float u[4][4];
+  // software pipeline: next step's 4x4 patch + 16 U elements load into
+  // registers UNDER the (long, 32-MFMA) compute phase; the transform + LDS
+  // stores run at the top of the next iteration
+  float d[4][4];
+  float ur[16];
+
+  auto load_regs = [&](int ci0) {
+    const float* xp = x + xplane + (long)ci0 * H * W;
+    #pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      const int ih = ih0 + a;
+      const bool hv = (unsigned)ih < (unsigned)H;
       #pragma unroll
       for (int bb = 0; bb < 4; ++bb) {
-        u[0][bb] = d[0][bb] - d[2][bb];
-        u[1][bb] = d[1][bb] + d[2][bb];
-        u[2][bb] = d[2][bb] - d[1][bb];
-        u[3][bb] = d[1][bb] - d[3][bb];
-      }
-      #pragma unroll
-      for (int a = 0; a < 4; ++a) {
-        const float v0 = u[a][0] - u[a][2];
-        const float v1 = u[a][1] + u[a][2];
-        const float v2 = u[a][2] - u[a][1];
-        const float v3 = u[a][1] - u[a][3];
-        Vl[((a * 4 + 0) * WF_CK + v_ci) * WF_VLD + v_t] = v0;
-        Vl[((a * 4 + 1) * WF_CK + v_ci) * WF_VLD + v_t] = v1;
-        Vl[((a * 4 + 2) * WF_CK + v_ci) * WF_VLD + v_t] = v2;
-        Vl[((a * 4 + 3) * WF_CK + v_ci) * WF_VLD + v_t] = v3;
+        const int iw = iw0 + bb;
+        const bool v = hv && (unsigned)iw < (unsigned)W;
+        d[a][bb] = v ? xp[(long)ih * W + iw] : 0.f;
       }
     }
-    // ---- U: 16 x 32 x 8 staged, 16 elements per thread
-    {
-      #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const int e = u_lin + j * 256;        // 0..4095
-        const int f = e >> 8;                 // 0..15
-        const int r = e & 255;                // co*8 + ci
-        const int co = r >> 3;
-        const int ci = r & 7;
-        Ul[(f * WF_CK + ci) * WF_ULD + co] =
-            U[((long)f * Co + co0 + co) * Ci + ci0 + ci];
-      }
+    #pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      const int e = u_lin + j * 256;        // 0..4095
+      const int f = e >> 8;                 // 0..15
+      const int r = e & 255;                // co*8 + ci
+      ur[j] = U[((long)f * Co + co0 + (r >> 3)) * Ci + ci0 + (r & 7)];
     }
   };
 
+  auto store_stage = [&]() {
+    float u[4][4];
+    #pragma unroll
+    for (int bb = 0; bb < 4; ++bb) {
+      u[0][bb] = d[0][bb] - d[2][bb];
+      u[1][bb] = d[1][bb] + d[2][bb];
+      u[2][bb] = d[2][bb] - d[1][bb];
+      u[3][bb] = d[1][bb] - d[3][bb];
+    }
+    #pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      Vl[((a * 4 + 0) * WF_CK + v_ci) * WF_VLD + v_t] = u[a][0] - u[a][2];
+      Vl[((a * 4 + 1) * WF_CK + v_ci) * WF_VLD + v_t] = u[a][1] + u[a][2];
+      Vl[((a * 4 + 2) * WF_CK + v_ci) * WF_VLD + v_t] = u[a][2] - u[a][1];
+      Vl[((a * 4 + 3) * WF_CK + v_ci) * WF_VLD + v_t] = u[a][1] - u[a][3];
+    }
+    #pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      const int e = u_lin + j * 256;
+      const int f = e >> 8;
+      const int r = e & 255;
+      Ul[(f * WF_CK + (r & 7)) * WF_ULD + (r >> 3)] = ur[j];
+    }
+  };
+
+  load_regs(0);
   for (int ci0 = 0; ci0 < Ci; ci0 += WF_CK) {
-    load_stage(ci0);
+    store_stage();
     __syncthreads();
+    if (ci0 + WF_CK < Ci) load_regs(ci0 + WF_CK);
     #pragma unroll
     for (int f = 0; f < 16; ++f) {
       #pragma unroll
