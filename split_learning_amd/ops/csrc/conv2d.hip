@@ -52,6 +52,24 @@ static inline bool conv_pad_mode() {
 at::Tensor conv2d_wino(const at::Tensor&, const at::Tensor&,
                        c10::optional<at::Tensor>, int, bool);
 at::Tensor conv2d_wino_bwdw(const at::Tensor&, const at::Tensor&, int);
+at::Tensor conv2d_wino_fused(const at::Tensor&, const at::Tensor&,
+                             c10::optional<at::Tensor>, int, bool);
+
+// fused-Winograd eligibility: shape constraints + grid fill.  The fused
+// kernel's grid is (T/32) x (Cout/32) with no ci-split, and the measured
+// win/lose line falls EXACTLY at >= 256 blocks (1 per CU): above it the
+// fused kernel beats every alternative (conv1_2 fwd 32.8 vs 44.3 us),
+// below it the chip underfills and it loses 2-4x (profiles/SUMMARY.md).
+static inline bool wino_fused_ok(int Cin, int Cout, int T, int OH, int OW) {
+  static int v = [] {
+    const char* e = std::getenv("SLK_WINO");
+    return e ? atoi(e) : 1;
+  }();
+  if (v == 0) return false;
+  if ((OH | OW) & 1) return false;
+  if (Cout % 32 != 0 || T % 32 != 0 || Cin % 8 != 0) return false;
+  return (long)(T / 32) * (Cout / 32) >= 256;
+}
 
 // Winograd routing (SLK_WINO=0 disables): F(2x2,3x3) beats the direct
 // implicit-GEMM kernel on the MEASURED win set only — square channel counts
@@ -573,11 +591,16 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                       bool x_is_padded) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.scalar_type() == at::kFloat);
   TORCH_CHECK(w.is_cuda() && w.dim() == 4 && w.size(1) == x.size(1));
-  if (!x_is_padded && !conv_pad_mode()
-      && wino_wins(x.size(1), w.size(0), x.size(2), w.size(2), w.size(3),
-                   stride, x.size(2) + 2 * pad - w.size(2) + 1,
-                   x.size(3) + 2 * pad - w.size(3) + 1)) {
-    return conv2d_wino(x, w, bias, pad, /*flip=*/false);
+  if (!x_is_padded && !conv_pad_mode() && stride == 1 && w.size(2) == 3
+      && w.size(3) == 3) {
+    const int OHw = x.size(2) + 2 * pad - 2, OWw = x.size(3) + 2 * pad - 2;
+    const int Tw = x.size(0) * (OHw / 2) * (OWw / 2);
+    if (wino_fused_ok(x.size(1), w.size(0), Tw, OHw, OWw)) {
+      return conv2d_wino_fused(x, w, bias, pad, /*flip=*/false);
+    }
+    if (wino_wins(x.size(1), w.size(0), x.size(2), 3, 3, 1, OHw, OWw)) {
+      return conv2d_wino(x, w, bias, pad, /*flip=*/false);
+    }
   }
   const bool padded = x_is_padded || conv_pad_mode();
   auto stream = c10::hip::getCurrentHIPStream().stream();
@@ -622,9 +645,14 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
   const int Ci = w.size(1), Co = w.size(0);
   const int B = gy.size(0);
   // gx = winograd-conv of gy with rotated/transposed weights at pad' = 2-p
-  if (stride == 1 && KH == 3 && KW == 3 && pad <= 2
-      && wino_wins(Co, Ci, gy.size(2), 3, 3, 1, H, W)) {
-    return conv2d_wino(gy, w, c10::nullopt, 2 - pad, /*flip=*/true);
+  if (stride == 1 && KH == 3 && KW == 3 && pad <= 2) {
+    const int Tw = B * (H / 2) * (W / 2);
+    if (H % 2 == 0 && W % 2 == 0 && wino_fused_ok(Co, Ci, Tw, H, W)) {
+      return conv2d_wino_fused(gy, w, c10::nullopt, 2 - pad, /*flip=*/true);
+    }
+    if (wino_wins(Co, Ci, gy.size(2), 3, 3, 1, H, W)) {
+      return conv2d_wino(gy, w, c10::nullopt, 2 - pad, /*flip=*/true);
+    }
   }
 
   if (conv_pad_mode() && stride == 1 && KH == KW && pad <= KH - 1) {
