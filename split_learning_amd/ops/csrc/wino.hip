@@ -310,11 +310,18 @@ constexpr int WF_CK = 8;    // ci step
 constexpr int WF_ULD = WF_CO + 1;   // U_lds[f][ci][co]
 constexpr int WF_VLD = WF_T + 1;    // V_lds[f][ci][t]
 
+// ci_per: the ci-range length per grid.z slice.  The output transform is
+// LINEAR in M, so slices transform their PARTIAL sums and atomicAdd into a
+// zeroed y — this fills the chip for small-T/high-Ci shapes whose natural
+// grid is far below one block per CU.  Slice 0 adds the bias.
 __global__ __launch_bounds__(256, 2) void wino_fused_kernel(
     const float* __restrict__ x, const float* __restrict__ U,
     const float* __restrict__ bias, float* __restrict__ y, int B, int Ci,
     int H, int W, int Co, int OH, int OW, int tH, int tW, int pad,
-    FastDiv d_thw, FastDiv d_tw) {
+    int ci_per, FastDiv d_thw, FastDiv d_tw) {
+  const int ci_begin = blockIdx.z * ci_per;
+  const int ci_end = min(Ci, ci_begin + ci_per);
+  const bool accumulate = gridDim.z > 1;
   // single-buffer two-barrier loop: double buffering would cost 67.6 KB of
   // LDS (2 blocks/CU); at 33.8 KB four blocks fit and the long 32-MFMA
   // phase gives co-resident waves the latency cover instead
@@ -348,7 +355,7 @@ __global__ __launch_bounds__(256, 2) void wino_fused_kernel(
   const unsigned vtw = d_tw.mod(vrem, vth);
   const int ih0 = (int)vth * 2 - pad;
   const int iw0 = (int)vtw * 2 - pad;
-  const long xplane = ((long)vb * Ci + v_ci) * H * W;  // + ciстеп*H*W later
+  const long xplane = ((long)vb * Ci + v_ci) * H * W;
   // U: each thread stages 16 elements (256 threads x 16 = 16f x 32co x 8ci)
   const int u_lin = tid;
 
@@ -405,11 +412,11 @@ __global__ __launch_bounds__(256, 2) void wino_fused_kernel(
     }
   };
 
-  load_regs(0);
-  for (int ci0 = 0; ci0 < Ci; ci0 += WF_CK) {
+  load_regs(ci_begin);
+  for (int ci0 = ci_begin; ci0 < ci_end; ci0 += WF_CK) {
     store_stage();
     __syncthreads();
-    if (ci0 + WF_CK < Ci) load_regs(ci0 + WF_CK);
+    if (ci0 + WF_CK < ci_end) load_regs(ci0 + WF_CK);
     #pragma unroll
     for (int f = 0; f < 16; ++f) {
       #pragma unroll
@@ -439,16 +446,27 @@ __global__ __launch_bounds__(256, 2) void wino_fused_kernel(
       u0[bb] = m[0][bb] + m[1][bb] + m[2][bb];
       u1[bb] = m[1][bb] - m[2][bb] - m[3][bb];
     }
-    const float bv = bias != nullptr ? bias[co] : 0.f;
+    const float bv = (bias != nullptr && blockIdx.z == 0) ? bias[co] : 0.f;
     const unsigned b = d_thw.div((unsigned)t);
     const unsigned rem = d_thw.mod((unsigned)t, b);
     const unsigned th = d_tw.div(rem);
     const unsigned tw = d_tw.mod(rem, th);
     float* yp = y + ((long)b * Co + co) * OH * OW + (long)th * 2 * OW + tw * 2;
-    yp[0] = u0[0] + u0[1] + u0[2] + bv;
-    yp[1] = u0[1] - u0[2] - u0[3] + bv;
-    yp[OW] = u1[0] + u1[1] + u1[2] + bv;
-    yp[OW + 1] = u1[1] - u1[2] - u1[3] + bv;
+    const float o00 = u0[0] + u0[1] + u0[2] + bv;
+    const float o01 = u0[1] - u0[2] - u0[3] + bv;
+    const float o10 = u1[0] + u1[1] + u1[2] + bv;
+    const float o11 = u1[1] - u1[2] - u1[3] + bv;
+    if (accumulate) {
+      atomicAdd(&yp[0], o00);
+      atomicAdd(&yp[1], o01);
+      atomicAdd(&yp[OW], o10);
+      atomicAdd(&yp[OW + 1], o11);
+    } else {
+      yp[0] = o00;
+      yp[1] = o01;
+      yp[OW] = o10;
+      yp[OW + 1] = o11;
+    }
   }
 }
 
@@ -533,16 +551,35 @@ at::Tensor conv2d_wino_fused(const at::Tensor& x, const at::Tensor& w,
                          (int)w.size(0), (int)w.size(1));
     }
   }
-  auto y = at::empty({B, Co, OH, OW}, x.options());
+  // ci-split so the grid reaches ~1 block/CU (output transform is linear in
+  // M: slices atomicAdd partial y tiles; slice count keeps ci chunks at
+  // multiples of the CK=8 step)
+  const long base_blocks = (long)(T / 32) * (Co / 32);
+  int splits = 1;
+  if (base_blocks < 256) {
+    splits = (int)((256 + base_blocks - 1) / base_blocks);
+    const int max_splits = Ci / WF_CK;
+    if (splits > max_splits) splits = max_splits;
+  }
+  int ci_per = ((Ci / splits + WF_CK - 1) / WF_CK) * WF_CK;
+  splits = (Ci + ci_per - 1) / ci_per;
+
+  at::Tensor y;
+  if (splits > 1) {
+    y = at::empty({B, Co, OH, OW}, x.options());
+    slk_zero_async(y.data_ptr<float>(), y.numel(), stream);
+  } else {
+    y = at::empty({B, Co, OH, OW}, x.options());
+  }
   FastDiv d_thw, d_tw;
   d_thw.init(tH * tW);
   d_tw.init(tW);
-  dim3 grid(T / 32, Co / 32);
+  dim3 grid(T / 32, Co / 32, splits);
   hipLaunchKernelGGL(wino_fused_kernel, grid, dim3(256), 0, stream,
                      xc.data_ptr<float>(), U.data_ptr<float>(),
                      bias.has_value() ? bias->data_ptr<float>() : nullptr,
                      y.data_ptr<float>(), B, Ci, H, W, Co, OH, OW, tH, tW,
-                     pad, d_thw, d_tw);
+                     pad, ci_per, d_thw, d_tw);
   return y;
 }
 
