@@ -29,42 +29,40 @@ namespace slk {
 template <bool FLIP>
 __global__ void wino_wt_kernel(const float* __restrict__ w,
                                float* __restrict__ U, int Cw0, int Cw1) {
+  // thread per (channel pair, frequency ROW): 4x the threads of the obvious
+  // per-pair mapping — the small conv weights (64x64: 4096 pairs) gave a
+  // 16-BLOCK grid on a 256-CU chip and the launch ran latency-bound at 7-9
+  // us; the 4x re-read of w is noise next to that
   const long total = (long)Cw0 * Cw1;
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
+  const long tstride = (long)gridDim.x * blockDim.x;
+  for (long e = (long)blockIdx.x * blockDim.x + threadIdx.x; e < 4 * total;
+       e += tstride) {
+    const int a = (int)(e & 3);          // frequency row
+    const long i = e >> 2;
     const int cols = FLIP ? Cw0 : Cw1;
     const int c = (int)(i % cols);
     const int r = (int)(i / cols);
     const int co = FLIP ? c : r;   // index into w's dim 0
     const int ci = FLIP ? r : c;   // index into w's dim 1
-    float g[3][3];
-    #pragma unroll
-    for (int a = 0; a < 3; ++a)
-      #pragma unroll
-      for (int b = 0; b < 3; ++b)
-        g[a][b] = w[(((long)co * Cw1 + ci) * 3 +
-                     (FLIP ? 2 - a : a)) * 3 + (FLIP ? 2 - b : b)];
-    // Gg: [4][3],  G = [[1,0,0],[.5,.5,.5],[.5,-.5,.5],[0,0,1]]
-    float t[4][3];
+    // row a of G g (G = [[1,0,0],[.5,.5,.5],[.5,-.5,.5],[0,0,1]])
+    float t[3];
     #pragma unroll
     for (int b = 0; b < 3; ++b) {
-      t[0][b] = g[0][b];
-      t[1][b] = 0.5f * (g[0][b] + g[1][b] + g[2][b]);
-      t[2][b] = 0.5f * (g[0][b] - g[1][b] + g[2][b]);
-      t[3][b] = g[2][b];
+      const float g0 = w[(((long)co * Cw1 + ci) * 3 + (FLIP ? 2 : 0)) * 3 +
+                         (FLIP ? 2 - b : b)];
+      const float g1 = w[(((long)co * Cw1 + ci) * 3 + 1) * 3 +
+                         (FLIP ? 2 - b : b)];
+      const float g2 = w[(((long)co * Cw1 + ci) * 3 + (FLIP ? 0 : 2)) * 3 +
+                         (FLIP ? 2 - b : b)];
+      t[b] = a == 0 ? g0
+           : a == 1 ? 0.5f * (g0 + g1 + g2)
+           : a == 2 ? 0.5f * (g0 - g1 + g2)
+           : g2;
     }
-    #pragma unroll
-    for (int a = 0; a < 4; ++a) {
-      const float u0 = t[a][0];
-      const float u1 = 0.5f * (t[a][0] + t[a][1] + t[a][2]);
-      const float u2 = 0.5f * (t[a][0] - t[a][1] + t[a][2]);
-      const float u3 = t[a][2];
-      U[((long)(a * 4 + 0) * total) + i] = u0;
-      U[((long)(a * 4 + 1) * total) + i] = u1;
-      U[((long)(a * 4 + 2) * total) + i] = u2;
-      U[((long)(a * 4 + 3) * total) + i] = u3;
-    }
+    U[((long)(a * 4 + 0) * total) + i] = t[0];
+    U[((long)(a * 4 + 1) * total) + i] = 0.5f * (t[0] + t[1] + t[2]);
+    U[((long)(a * 4 + 2) * total) + i] = 0.5f * (t[0] - t[1] + t[2]);
+    U[((long)(a * 4 + 3) * total) + i] = t[2];
   }
 }
 
@@ -252,7 +250,7 @@ at::Tensor conv2d_wino(const at::Tensor& x, const at::Tensor& w,
   auto U = at::empty({16, Co, Ci}, x.options());
   auto V = at::empty({16, Ci, T}, x.options());
   {
-    const long tot = (long)Co * Ci;
+    const long tot = (long)Co * Ci * 4;   // thread per (pair, freq row)
     const int grid = (int)std::min<long>((tot + 255) / 256, 4096);
     // always the WEIGHT's own dims; FLIP handles the transpose internally
     if (flip) {
@@ -539,7 +537,7 @@ at::Tensor conv2d_wino_fused(const at::Tensor& x, const at::Tensor& w,
 
   auto U = at::empty({16, Co, Ci}, x.options());
   {
-    const long tot = (long)Co * Ci;
+    const long tot = (long)Co * Ci * 4;
     const int grid = (int)std::min<long>((tot + 255) / 256, 4096);
     if (flip) {
       hipLaunchKernelGGL(wino_wt_kernel<true>, dim3(grid), dim3(256), 0,

@@ -623,3 +623,26 @@ def test_conv_bias_grad_zero_under_bn():
     xr = x.clone()
     ref(xr).sum().backward()
     assert float(ref[0].bias.grad.abs().max()) < 1e-3
+
+
+@pytest.mark.gpu
+def test_conv2d_wino_fused_direct_and_split():
+    """Fused Winograd kernel called directly: the big-T routed shape AND a
+    small-T shape that exercises the ci-split (atomicAdd partial y) path."""
+    from split_learning_amd.ops.functional import native
+    n = native()
+    torch.manual_seed(13)
+    for (B, Ci, H, Co) in [(32, 64, 32, 64), (8, 512, 4, 512)]:
+        x = torch.randn(B, Ci, H, H, device="cuda")
+        w = torch.randn(Co, Ci, 3, 3, device="cuda") * 0.1
+        b = torch.randn(Co, device="cuda")
+        y = n.conv2d_wino_fused(x, w, b, 1, False)
+        ref = F.conv2d(x, w, b, 1, 1)
+        assert_close(y, ref, atol=2e-4, rtol=2e-4,
+                     what=f"wino_fused {Ci}x{H}->{Co}")
+        gy = torch.randn(B, Co, H, H, device="cuda")
+        gx = n.conv2d_wino_fused(gy, w, None, 1, True)
+        xr = x.clone().requires_grad_(True)
+        F.conv2d(xr, w, None, 1, 1).backward(gy)
+        assert_close(gx, xr.grad, atol=2e-4, rtol=2e-4,
+                     what=f"wino_fused bwdd {Ci}x{H}->{Co}")
