@@ -13,6 +13,7 @@ at::Tensor conv2d_wino(const at::Tensor&, const at::Tensor&,
 at::Tensor conv2d_wino_bwdw(const at::Tensor&, const at::Tensor&, int);
 at::Tensor conv2d_wino_fused(const at::Tensor&, const at::Tensor&,
                              c10::optional<at::Tensor>, int, bool);
+at::Tensor conv2d_wino_bwdw_fused(const at::Tensor&, const at::Tensor&, int);
 // conv2d.hip
 at::Tensor pad_nchw(const at::Tensor&, int);
 at::Tensor conv2d_fwd(const at::Tensor&, const at::Tensor&, c10::optional<at::Tensor>,
@@ -96,6 +97,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_wino_fused", &slk::conv2d_wino_fused, py::arg("x"),
         py::arg("w"), py::arg("bias") = py::none(), py::arg("pad") = 1,
         py::arg("flip") = false);
+  m.def("conv2d_wino_bwdw_fused", &slk::conv2d_wino_bwdw_fused, py::arg("gy"),
+        py::arg("x"), py::arg("pad") = 1);
   m.def("pad_nchw", &slk::pad_nchw);
   m.def("conv2d_fwd", &slk::conv2d_fwd, py::arg("x"), py::arg("w"),
         py::arg("bias"), py::arg("stride"), py::arg("pad"),

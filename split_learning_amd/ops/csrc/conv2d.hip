@@ -54,6 +54,7 @@ at::Tensor conv2d_wino(const at::Tensor&, const at::Tensor&,
 at::Tensor conv2d_wino_bwdw(const at::Tensor&, const at::Tensor&, int);
 at::Tensor conv2d_wino_fused(const at::Tensor&, const at::Tensor&,
                              c10::optional<at::Tensor>, int, bool);
+at::Tensor conv2d_wino_bwdw_fused(const at::Tensor&, const at::Tensor&, int);
 
 // fused-Winograd eligibility: shape constraints + grid fill.  The fused
 // kernel's grid is (T/32) x (Cout/32) with no ci-split, and the measured
@@ -701,16 +702,25 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
 at::Tensor conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x, int KH,
                              int KW, int stride, int pad, bool x_is_padded) {
   TORCH_CHECK(gy.is_cuda() && x.is_cuda() && gy.scalar_type() == at::kFloat);
-  // Winograd weight-gradient: dU[f] = (A gy A^T)[f] @ V_x[f]^T then
-  // gw = G^T dU G — replaces the direct gather's pathological tall-K GEMM
-  // ([Co, Ci*9, B*OH*OW]: 9 n-tiles at conv1_2) with 16 well-shaped
-  // frequency GEMMs.  Same routing guard as fwd (bwd-weight win set is
-  // broader; SLK_WINO applies).
+  // Winograd weight-gradient.  FUSED twin first (in-kernel transforms,
+  // tile-axis reduction with per-slice slabs): measured faster than every
+  // alternative on ALL eligible shapes — 512ch 2x2: 13.2 us vs 29.1 routed,
+  // 512ch 4x4: 36.2 vs 48.2, 64->128: 33.5 vs 45.9 (profiles/SUMMARY.md).
+  // The transform+frequency-GEMM pipeline (conv2d_wino_bwdw) remains for
+  // Ci/Co not multiples of 32.
   if (!x_is_padded && !conv_pad_mode() && stride == 1 && KH == 3 && KW == 3
-      && pad <= 2
-      && wino_wins_bwdw(x.size(1), gy.size(1), x.size(2), gy.size(2),
-                        gy.size(3))) {
-    return conv2d_wino_bwdw(gy, x, pad);
+      && pad <= 2 && gy.size(2) % 2 == 0 && gy.size(3) % 2 == 0) {
+    static int wv = [] {
+      const char* e = std::getenv("SLK_WINO");
+      return e ? atoi(e) : 1;
+    }();
+    if (wv != 0 && x.size(1) % 32 == 0 && gy.size(1) % 32 == 0) {
+      return conv2d_wino_bwdw_fused(gy, x, pad);
+    }
+    if (wino_wins_bwdw(x.size(1), gy.size(1), x.size(2), gy.size(2),
+                       gy.size(3))) {
+      return conv2d_wino_bwdw(gy, x, pad);
+    }
   }
   const bool padded = x_is_padded || conv_pad_mode();
   auto gyc = gy.contiguous();

@@ -514,6 +514,217 @@ at::Tensor conv2d_wino_bwdw(const at::Tensor& gy, const at::Tensor& x,
   return gw;
 }
 
+// ---- fused backward-weight --------------------------------------------
+// Twin of wino_fused_kernel with the TILE axis as the reduction: block owns
+// a (32co x 32ci) patch of gw across all 16 frequencies, K-loop over tiles
+// staging (A gy A^T) and (B^T x B) patches transformed in-register, epilogue
+// applies G^T dU G.  grid.z slices the tile range; slices write per-split
+// slabs (non-atomic) reduced by slab_reduce (the 2.3M-atomicAdd alternative
+// re-creates the split-K atomic pathology measured in round 2).
+__global__ __launch_bounds__(256, 2) void wino_bwdw_fused_kernel(
+    const float* __restrict__ gy, const float* __restrict__ x,
+    float* __restrict__ slab, int B, int Ci, int H, int W, int Co, int OH,
+    int OW, int tH, int tW, int pad, int t_per, FastDiv d_thw, FastDiv d_tw) {
+  __shared__ float Gl[16 * 8 * 33];   // [f][tloc][co]
+  __shared__ float Xl[16 * 8 * 33];   // [f][tloc][ci]
+
+  const int T = B * tH * tW;
+  const int ci0 = blockIdx.y * 32;
+  const int co0 = blockIdx.x * 32;
+  const int t_begin = blockIdx.z * t_per;
+  const int t_end = min(T, t_begin + t_per);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wco = (wid >> 1) * 16;
+  const int wci = (wid & 1) * 16;
+  const int frag_r = lane >> 4;
+  const int frag_c = lane & 15;
+
+  f32x4 acc[16];
+  #pragma unroll
+  for (int f = 0; f < 16; ++f) acc[f] = f32x4{};
+
+  // staging: thread -> (channel = tid>>3, tloc = tid&7) for BOTH sides
+  const int s_ch = tid >> 3;          // 0..31
+  const int s_t = tid & 7;            // 0..7
+
+  float gg[2][2];     // gy 2x2 tile (for co0 + s_ch)
+  float xd[4][4];     // x 4x4 patch (for ci0 + s_ch)
+
+  auto load_regs = [&](int t0) {
+    const int t = t0 + s_t;
+    const int tc = t < T ? t : T - 1;
+    const bool tv = t < t_end;
+    const unsigned b = d_thw.div((unsigned)tc);
+    const unsigned rem = d_thw.mod((unsigned)tc, b);
+    const unsigned th = d_tw.div(rem);
+    const unsigned tw = d_tw.mod(rem, th);
+    {
+      const float* gp = gy + ((long)b * Co + co0 + s_ch) * OH * OW
+                        + (long)th * 2 * OW + tw * 2;
+      gg[0][0] = tv ? gp[0] : 0.f;
+      gg[0][1] = tv ? gp[1] : 0.f;
+      gg[1][0] = tv ? gp[OW] : 0.f;
+      gg[1][1] = tv ? gp[OW + 1] : 0.f;
+    }
+    {
+      const int ih0 = (int)th * 2 - pad;
+      const int iw0 = (int)tw * 2 - pad;
+      const float* xp = x + ((long)b * Ci + ci0 + s_ch) * H * W;
+      #pragma unroll
+      for (int a = 0; a < 4; ++a) {
+        const int ih = ih0 + a;
+        const bool hv = tv && (unsigned)ih < (unsigned)H;
+        #pragma unroll
+        for (int bb = 0; bb < 4; ++bb) {
+          const int iw = iw0 + bb;
+          const bool v = hv && (unsigned)iw < (unsigned)W;
+          xd[a][bb] = v ? xp[(long)ih * W + iw] : 0.f;
+        }
+      }
+    }
+  };
+
+  auto store_stage = [&]() {
+    // dM = A gy A^T (A = [[1,0],[1,1],[1,-1],[0,-1]])
+    const float r0c0 = gg[0][0], r0c1 = gg[0][1];
+    const float r1c0 = gg[0][0] + gg[1][0], r1c1 = gg[0][1] + gg[1][1];
+    const float r2c0 = gg[0][0] - gg[1][0], r2c1 = gg[0][1] - gg[1][1];
+    const float r3c0 = -gg[1][0], r3c1 = -gg[1][1];
+    #pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      const float c0 = a == 0 ? r0c0 : a == 1 ? r1c0 : a == 2 ? r2c0 : r3c0;
+      const float c1 = a == 0 ? r0c1 : a == 1 ? r1c1 : a == 2 ? r2c1 : r3c1;
+      Gl[((a * 4 + 0) * 8 + s_t) * 33 + s_ch] = c0;
+      Gl[((a * 4 + 1) * 8 + s_t) * 33 + s_ch] = c0 + c1;
+      Gl[((a * 4 + 2) * 8 + s_t) * 33 + s_ch] = c0 - c1;
+      Gl[((a * 4 + 3) * 8 + s_t) * 33 + s_ch] = -c1;
+    }
+    // B^T d B
+    float u[4][4];
+    #pragma unroll
+    for (int bb = 0; bb < 4; ++bb) {
+      u[0][bb] = xd[0][bb] - xd[2][bb];
+      u[1][bb] = xd[1][bb] + xd[2][bb];
+      u[2][bb] = xd[2][bb] - xd[1][bb];
+      u[3][bb] = xd[1][bb] - xd[3][bb];
+    }
+    #pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      Xl[((a * 4 + 0) * 8 + s_t) * 33 + s_ch] = u[a][0] - u[a][2];
+      Xl[((a * 4 + 1) * 8 + s_t) * 33 + s_ch] = u[a][1] + u[a][2];
+      Xl[((a * 4 + 2) * 8 + s_t) * 33 + s_ch] = u[a][2] - u[a][1];
+      Xl[((a * 4 + 3) * 8 + s_t) * 33 + s_ch] = u[a][1] - u[a][3];
+    }
+  };
+
+  load_regs(t_begin);
+  for (int t0 = t_begin; t0 < t_end; t0 += 8) {
+    store_stage();
+    __syncthreads();
+    if (t0 + 8 < t_end) load_regs(t0 + 8);
+    #pragma unroll
+    for (int f = 0; f < 16; ++f) {
+      #pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int kr = kk * 4 + frag_r;
+        const float a = Gl[(f * 8 + kr) * 33 + wco + frag_c];
+        const float b = Xl[(f * 8 + kr) * 33 + wci + frag_c];
+        acc[f] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[f], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: lane holds per freq rows co = wco + frag_r*4 + i, col ci =
+  // wci + frag_c; gw = G^T dU G per (co, ci), written to this slice's slab
+  const long numel = (long)Co * Ci * 9;
+  #pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int co = co0 + wco + frag_r * 4 + i;
+    const int ci = ci0 + wci + frag_c;
+    if (co >= Co || ci >= Ci) continue;
+    float u[4][4];
+    #pragma unroll
+    for (int f = 0; f < 16; ++f) u[f >> 2][f & 3] = acc[f][i];
+    float t3[3][4];
+    #pragma unroll
+    for (int bb = 0; bb < 4; ++bb) {
+      t3[0][bb] = u[0][bb] + 0.5f * (u[1][bb] + u[2][bb]);
+      t3[1][bb] = 0.5f * (u[1][bb] - u[2][bb]);
+      t3[2][bb] = 0.5f * (u[1][bb] + u[2][bb]) + u[3][bb];
+    }
+    float* gp = slab + (long)blockIdx.z * numel + ((long)co * Ci + ci) * 9;
+    #pragma unroll
+    for (int r = 0; r < 3; ++r) {
+      gp[r * 3 + 0] = t3[r][0] + 0.5f * (t3[r][1] + t3[r][2]);
+      gp[r * 3 + 1] = 0.5f * (t3[r][1] - t3[r][2]);
+      gp[r * 3 + 2] = 0.5f * (t3[r][1] + t3[r][2]) + t3[r][3];
+    }
+  }
+}
+
+__global__ void wino_slab_reduce_kernel(const float* __restrict__ slab,
+                                        int splits, float* __restrict__ out,
+                                        long numel) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < numel;
+       i += stride) {
+    float s = 0.f;
+    for (int k = 0; k < splits; ++k) s += slab[(long)k * numel + i];
+    out[i] = s;
+  }
+}
+
+at::Tensor conv2d_wino_bwdw_fused(const at::Tensor& gy, const at::Tensor& x,
+                                  int pad) {
+  TORCH_CHECK(gy.is_cuda() && x.is_cuda() && gy.dim() == 4 && x.dim() == 4);
+  auto gyc = gy.contiguous();
+  auto xc = x.contiguous();
+  const int B = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int Co = gy.size(1), OH = gy.size(2), OW = gy.size(3);
+  TORCH_CHECK(OH == H + 2 * pad - 2 && OW == W + 2 * pad - 2,
+              "wino_bwdw_fused geometry");
+  TORCH_CHECK(OH % 2 == 0 && OW % 2 == 0 && Co % 32 == 0 && Ci % 32 == 0,
+              "wino_bwdw_fused shape requirements");
+  const int tH = OH / 2, tW = OW / 2;
+  const int T = B * tH * tW;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+
+  const long base_blocks = (long)(Co / 32) * (Ci / 32);
+  int splits = 1;
+  if (base_blocks < 256) {
+    splits = (int)((256 + base_blocks - 1) / base_blocks);
+    const int max_splits = (T + 7) / 8;
+    if (splits > max_splits) splits = max_splits;
+  }
+  int t_per = ((T / splits + 7) / 8) * 8;
+  if (t_per < 8) t_per = 8;
+  splits = (T + t_per - 1) / t_per;
+
+  auto slab = at::empty({splits, Co, Ci, 3, 3}, x.options());
+  FastDiv d_thw, d_tw;
+  d_thw.init(tH * tW);
+  d_tw.init(tW);
+  dim3 grid(Co / 32, Ci / 32, splits);
+  hipLaunchKernelGGL(wino_bwdw_fused_kernel, grid, dim3(256), 0, stream,
+                     gyc.data_ptr<float>(), xc.data_ptr<float>(),
+                     slab.data_ptr<float>(), B, Ci, H, W, Co, OH, OW, tH, tW,
+                     pad, t_per, d_thw, d_tw);
+  auto gw = at::empty({Co, Ci, 3, 3}, x.options());
+  if (splits == 1) {
+    return slab.view({Co, Ci, 3, 3});
+  }
+  const long numel = (long)Co * Ci * 9;
+  const int rgrid = (int)std::min<long>((numel + 255) / 256, 4096);
+  hipLaunchKernelGGL(wino_slab_reduce_kernel, dim3(rgrid), dim3(256), 0,
+                     stream, slab.data_ptr<float>(), splits,
+                     gw.data_ptr<float>(), numel);
+  return gw;
+}
+
 // fully-fused variant: transforms in-kernel, no V/M round trip.  Shape
 // requirements: Co % 32 == 0, T % 32 == 0, Ci % 8 == 0.  SLK_WINO_FUSED=0
 // falls back to the unfused pipeline (A/B).
