@@ -31,7 +31,7 @@ static inline at::Tensor zeroed(at::IntArrayRef sizes, const at::TensorOptions& 
 // atomics); the finalize kernel reduces over chunks
 __global__ void bn_partial_kernel(const float* __restrict__ x,
                                   float* __restrict__ slab,
-                                  int B, int C, int HW) {
+                                  int B, int C, int HW, FastDiv d_hw) {
   __shared__ double scratch[16];
   const int c = blockIdx.x;
   const int total = B * HW;
@@ -40,8 +40,9 @@ __global__ void bn_partial_kernel(const float* __restrict__ x,
   const int hi = min(total, lo + per);
   double s = 0.0, s2 = 0.0;
   for (int i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-    const int b = i / HW;
-    const int r = i - b * HW;
+    // FastDiv: the plain i / HW was a ~25-VALU runtime division PER ELEMENT
+    const unsigned b = d_hw.div((unsigned)i);
+    const unsigned r = d_hw.mod((unsigned)i, b);
     const double v = (double)x[((long)b * C + c) * HW + r];
     s += v;
     s2 += v * v;
@@ -88,10 +89,11 @@ __global__ void bn_fwd_kernel(const float* __restrict__ x, float* __restrict__ y
                               const float* __restrict__ invstd,
                               const float* __restrict__ gamma,
                               const float* __restrict__ beta, long total, int C,
-                              int HW, bool relu) {
+                              int HW, bool relu, FastDiv d_hw, FastDiv d_c) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
-    const int c = (int)((i / HW) % C);
+    const unsigned bc = d_hw.div((unsigned)i);
+    const int c = (int)(bc - d_c.div(bc) * (unsigned)C);
     float v = (x[i] - mean[c]) * invstd[c] * gamma[c] + beta[c];
     if (relu) v = fmaxf(v, 0.f);
     y[i] = v;
@@ -110,7 +112,7 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
                                      float* __restrict__ slab, int B, int C,
-                                     int HW) {
+                                     int HW, FastDiv d_hw) {
   __shared__ double scratch[16];
   const int c = blockIdx.x;
   const float m = mean[c], is = invstd[c];
@@ -120,8 +122,8 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
   const int hi = min(total, lo + per);
   double s = 0.0, sx = 0.0;
   for (int i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-    const int b = i / HW;
-    const int r = i - b * HW;
+    const unsigned b = d_hw.div((unsigned)i);
+    const unsigned r = d_hw.mod((unsigned)i, b);
     const long off = ((long)b * C + c) * HW + r;
     float gf = gy[off];
     if (relu_y != nullptr && relu_y[off] <= 0.f) gf = 0.f;
@@ -162,10 +164,12 @@ __global__ void bn_bwd_dx_kernel(const float* __restrict__ x,
                                  const float* __restrict__ sum_gy,
                                  const float* __restrict__ sum_gy_xhat,
                                  float* __restrict__ gx, long total, int C, int HW,
-                                 float inv_n, bool training) {
+                                 float inv_n, bool training, FastDiv d_hw,
+                                 FastDiv d_c) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
-    const int c = (int)((i / HW) % C);
+    const unsigned bc = d_hw.div((unsigned)i);
+    const int c = (int)(bc - d_c.div(bc) * (unsigned)C);
     const float is = invstd[c];
     float g = gy[i];
     if (relu_y != nullptr && relu_y[i] <= 0.f) g = 0.f;
@@ -197,8 +201,11 @@ std::vector<at::Tensor> bn2d_stats_fused(const at::Tensor& x,
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int chunks = bn_chunks((long)B * HW);
   auto slab = at::empty({2, chunks, C}, x.options());
+  FastDiv d_hw;
+  d_hw.init(HW);
   hipLaunchKernelGGL(bn_partial_kernel, dim3(C, chunks), dim3(256), 0, stream,
-                     x.data_ptr<float>(), slab.data_ptr<float>(), B, C, HW);
+                     x.data_ptr<float>(), slab.data_ptr<float>(), B, C, HW,
+                     d_hw);
   hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
                      stream, slab.data_ptr<float>(), chunks,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
@@ -220,11 +227,15 @@ at::Tensor bn2d_fwd(const at::Tensor& x, const at::Tensor& mean,
   const int C = x.size(1), HW = x.size(2) * x.size(3);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   int grid = (int)std::min<long>((total + 255) / 256, 2048);
+  FastDiv d_hw, d_c;
+  d_hw.init(HW);
+  d_c.init(C);
+  TORCH_CHECK(total < (1 << 24), "bn_fwd: FastDiv range");
   hipLaunchKernelGGL(bn_fwd_kernel, dim3(grid), dim3(256), 0, stream,
                      x.data_ptr<float>(), y.data_ptr<float>(),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      gamma.data_ptr<float>(), beta.data_ptr<float>(), total, C, HW,
-                     relu);
+                     relu, d_hw, d_c);
   return y;
 }
 
@@ -242,21 +253,26 @@ static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tens
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int rchunks = bn_chunks((long)B * HW);
   auto slab = at::empty({2, rchunks, C}, x.options());
+  FastDiv d_hw;
+  d_hw.init(HW);
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C, rchunks),
                      dim3(256), 0, stream,
                      x.data_ptr<float>(), gy.data_ptr<float>(), ry,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     slab.data_ptr<float>(), B, C, HW);
+                     slab.data_ptr<float>(), B, C, HW, d_hw);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
                      stream, slab.data_ptr<float>(), rchunks,
                      sum_gy.data_ptr<float>(), sum_gy_xhat.data_ptr<float>(), C);
   int grid = (int)std::min<long>((total + 255) / 256, 2048);
+  FastDiv d_c;
+  d_c.init(C);
+  TORCH_CHECK(total < (1 << 24), "bn_bwd: FastDiv range");
   hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(grid), dim3(256), 0, stream,
                      x.data_ptr<float>(), gy.data_ptr<float>(), ry,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      gamma.data_ptr<float>(), sum_gy.data_ptr<float>(),
                      sum_gy_xhat.data_ptr<float>(), gx.data_ptr<float>(), total, C,
-                     HW, 1.0f / (float)((long)B * HW), training);
+                     HW, 1.0f / (float)((long)B * HW), training, d_hw, d_c);
   // ggamma = sum_gy_xhat, gbeta = sum_gy
   return {gx, sum_gy_xhat, sum_gy};
 }
