@@ -210,14 +210,16 @@ at::Tensor dropout_bwd(const at::Tensor& gy, const at::Tensor& mask, double p) {
 // ---------------- MaxPool2d 2x2 s2 with argmax stash ----------------
 __global__ void maxpool_fwd_kernel(const float* __restrict__ x, float* __restrict__ y,
                                    uint8_t* __restrict__ idx, int BC, int H, int W,
-                                   int OH, int OW) {
+                                   int OH, int OW, FastDiv d_ohow, FastDiv d_ow) {
   const long total = (long)BC * OH * OW;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
-    const int ow = (int)(i % OW);
-    const int oh = (int)((i / OW) % OH);
-    const long bc = i / ((long)OH * OW);
-    const float* xp = x + (bc * H + oh * 2) * W + ow * 2;
+    // FastDiv: plain runtime / and % were ~25-VALU divisions per element
+    const unsigned bc = d_ohow.div((unsigned)i);
+    const unsigned rem = d_ohow.mod((unsigned)i, bc);
+    const unsigned oh = d_ow.div(rem);
+    const unsigned ow = d_ow.mod(rem, oh);
+    const float* xp = x + ((long)bc * H + oh * 2) * W + ow * 2;
     float best = xp[0];
     int bi = 0;
     if (ow * 2 + 1 < W && xp[1] > best) { best = xp[1]; bi = 1; }
@@ -233,17 +235,18 @@ __global__ void maxpool_fwd_kernel(const float* __restrict__ x, float* __restric
 __global__ void maxpool_bwd_kernel(const float* __restrict__ gy,
                                    const uint8_t* __restrict__ idx,
                                    float* __restrict__ gx, int BC, int H, int W,
-                                   int OH, int OW) {
+                                   int OH, int OW, FastDiv d_hw, FastDiv d_w) {
   const long total = (long)BC * H * W;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
-    const int w = (int)(i % W);
-    const int h = (int)((i / W) % H);
-    const long bc = i / ((long)H * W);
+    const unsigned bc = d_hw.div((unsigned)i);
+    const unsigned rem = d_hw.mod((unsigned)i, bc);
+    const unsigned h = d_w.div(rem);
+    const unsigned w = d_w.mod(rem, h);
     const int oh = h >> 1, ow = w >> 1;
     float g = 0.f;
     if (oh < OH && ow < OW) {
-      const long o = (bc * OH + oh) * OW + ow;
+      const long o = ((long)bc * OH + oh) * OW + ow;
       const int pos = ((h & 1) << 1) | (w & 1);
       if (idx[o] == (uint8_t)pos) g = gy[o];
     }
@@ -258,9 +261,12 @@ std::vector<at::Tensor> maxpool2x2_fwd(const at::Tensor& x) {
   auto idx = at::empty({B, C, OH, OW}, x.options().dtype(at::kByte));
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const long total = (long)B * C * OH * OW;
+  FastDiv d_ohow, d_ow;
+  d_ohow.init(OH * OW);
+  d_ow.init(OW);
   hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(ew_grid(total, 256, 1)), dim3(256), 0,
                      stream, x.data_ptr<float>(), y.data_ptr<float>(),
-                     idx.data_ptr<uint8_t>(), B * C, H, W, OH, OW);
+                     idx.data_ptr<uint8_t>(), B * C, H, W, OH, OW, d_ohow, d_ow);
   return {y, idx};
 }
 
@@ -269,9 +275,12 @@ at::Tensor maxpool2x2_bwd(const at::Tensor& gy, const at::Tensor& idx, int H, in
   auto gx = at::empty({B, C, H, W}, gy.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const long total = (long)B * C * H * W;
+  FastDiv d_hw, d_w;
+  d_hw.init(H * W);
+  d_w.init(W);
   hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(ew_grid(total, 256, 1)), dim3(256), 0,
                      stream, gy.data_ptr<float>(), idx.data_ptr<uint8_t>(),
-                     gx.data_ptr<float>(), B * C, H, W, OH, OW);
+                     gx.data_ptr<float>(), B * C, H, W, OH, OW, d_hw, d_w);
   return gx;
 }
 
