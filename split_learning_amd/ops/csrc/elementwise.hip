@@ -264,6 +264,7 @@ std::vector<at::Tensor> maxpool2x2_fwd(const at::Tensor& x) {
   FastDiv d_ohow, d_ow;
   d_ohow.init(OH * OW);
   d_ow.init(OW);
+  TORCH_CHECK(total <= (1l << 24), "maxpool_fwd: FastDiv range");
   hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(ew_grid(total, 256, 1)), dim3(256), 0,
                      stream, x.data_ptr<float>(), y.data_ptr<float>(),
                      idx.data_ptr<uint8_t>(), B * C, H, W, OH, OW, d_ohow, d_ow);
@@ -278,6 +279,7 @@ at::Tensor maxpool2x2_bwd(const at::Tensor& gy, const at::Tensor& idx, int H, in
   FastDiv d_hw, d_w;
   d_hw.init(H * W);
   d_w.init(W);
+  TORCH_CHECK(total <= (1l << 24), "maxpool_bwd: FastDiv range");
   hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(ew_grid(total, 256, 1)), dim3(256), 0,
                      stream, gy.data_ptr<float>(), idx.data_ptr<uint8_t>(),
                      gx.data_ptr<float>(), B * C, H, W, OH, OW, d_hw, d_w);

@@ -230,7 +230,7 @@ at::Tensor bn2d_fwd(const at::Tensor& x, const at::Tensor& mean,
   FastDiv d_hw, d_c;
   d_hw.init(HW);
   d_c.init(C);
-  TORCH_CHECK(total < (1 << 24), "bn_fwd: FastDiv range");
+  TORCH_CHECK(total <= (1l << 24), "bn_fwd: FastDiv range");
   hipLaunchKernelGGL(bn_fwd_kernel, dim3(grid), dim3(256), 0, stream,
                      x.data_ptr<float>(), y.data_ptr<float>(),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
@@ -266,7 +266,7 @@ static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tens
   int grid = (int)std::min<long>((total + 255) / 256, 2048);
   FastDiv d_c;
   d_c.init(C);
-  TORCH_CHECK(total < (1 << 24), "bn_bwd: FastDiv range");
+  TORCH_CHECK(total <= (1l << 24), "bn_bwd: FastDiv range");
   hipLaunchKernelGGL(bn_bwd_dx_kernel, dim3(grid), dim3(256), 0, stream,
                      x.data_ptr<float>(), gy.data_ptr<float>(), ry,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
