@@ -1,4 +1,7 @@
 // Winograd F(2x2, 3x3) convolution for stride-1 3x3 convs (fp32).
+// Op sites replaced: every 3x3 s1 Conv2d of the partitioned model zoo
+// (reference src/model/VGG16_CIFAR10.py:10-94,
+// other/Vanilla_SL/src/model/MobileNetv1_CIFAR10.py:23-167).
 //
 // y = A^T [ (G g G^T) .* (B^T d B) ] A   per 2x2 output tile (4x4 input
 // patch, overlap 2).  2.25x fewer MACs than direct: the elementwise product
