@@ -16,8 +16,11 @@
 //
 // Backward-data reuses the whole machinery: gx = winograd-conv of gy with
 // the 180-degree-rotated, Co/Ci-transposed weights at pad' = KH-1-p.
-// Backward-weight stays on the direct gather (a correlation, not a 3x3
-// conv over the same axes).
+// Backward-weight has its own frequency form (dU[f] = (A gy A^T)[f] @
+// V_x[f]^T, gw = G^T dU G) in both a transform+GEMM pipeline and a fully
+// FUSED kernel (wino_bwdw_fused_kernel).  The fully fused forward
+// (wino_fused_kernel) keeps the transforms in-kernel; routing between the
+// three forms is measured per shape (conv2d.hip).
 #include <torch/extension.h>
 #include <ATen/ATen.h>
 #include <c10/hip/HIPStream.h>
