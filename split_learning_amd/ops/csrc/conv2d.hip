@@ -87,8 +87,10 @@ static inline bool wino_wins_bwdw(int Ci, int Co, int H, int OH, int OW) {
   if (v >= 2) return true;
   // measured win set (tools/wino_check.py under SLK_WINO=0): square channel
   // counts win everywhere EXCEPT the 4x4 layers, where T = B*4 makes the
-  // frequency GEMMs too small (512ch 4x4: 62 vs 48 us direct)
-  return Ci == Co && H != 4;
+  // frequency GEMMs too small (512ch 4x4: 62 vs 48 us direct); the stem
+  // (3ch 32x32) also wins mildly (44.9 vs 49.3 us) — its direct form is a
+  // 9-n-tile tall-K pathology
+  return (Ci == Co && H != 4) || (Ci <= 4 && H >= 32);
 }
 
 static inline bool wino_wins(int Ci, int Co, int H, int KH, int KW, int stride,
