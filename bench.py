@@ -124,15 +124,29 @@ class ColocatedPipeline:
         self.s1_opt.step()
 
     def _capture(self):
-        # stable grad/descriptor pointers are required under capture
-        self.s1_opt.release_grads = False
-        self.s2_opt.release_grads = False
+        # Grads are None entering capture (release_grads default): inside the
+        # captured step AccumulateGrad STEALS each backward kernel's output
+        # tensor instead of add_-ing into a persistent buffer, removing one
+        # elementwise add per parameter per step (~45 launches, was 9% of
+        # kernel time).  The capture-pool grad buffers stay stable across
+        # replays and every gradient producer fully overwrites its output, so
+        # replays stay correct; the optimizer descriptor is rebuilt inside
+        # capture via a pinned-memory memcpy node (optim._build_desc).
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(3):
                 self._step()
         torch.cuda.current_stream().wait_stream(s)
+        # Flip release AFTER warmup (grads are None now, so the captured step
+        # still steals) but BEFORE capture: the captured step must NOT free
+        # the stolen grad buffers mid-capture — a grad block freed inside
+        # capture can be re-handed to a later allocation while a still-later
+        # graph node (the fused optimizer) reads it, which intermittently
+        # faulted on replay.  With release off, the params keep the pool
+        # buffers alive for the graph's lifetime.
+        self.s1_opt.release_grads = False
+        self.s2_opt.release_grads = False
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             self._step()

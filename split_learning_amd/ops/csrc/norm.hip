@@ -84,6 +84,47 @@ __global__ void bn_finalize_kernel(const float* __restrict__ slab, int chunks,
   }
 }
 
+// chunks==1 (B*HW <= 8191: the 8x8-and-smaller VGG layers at B=32): ONE
+// block per channel does the whole reduction AND the finalize — no slab
+// tensor, no second launch.  Same math as partial+finalize (the double-
+// precision block sum feeds the finalize directly).
+__global__ void bn_stats_one_kernel(const float* __restrict__ x,
+                                    float* __restrict__ mean,
+                                    float* __restrict__ invstd,
+                                    float* __restrict__ running_mean,
+                                    float* __restrict__ running_var,
+                                    long* __restrict__ nbt, int B, int C,
+                                    int HW, float momentum, float eps,
+                                    FastDiv d_hw) {
+  __shared__ double scratch[16];
+  const int c = blockIdx.x;
+  if (c == 0 && threadIdx.x == 0 && nbt != nullptr) nbt[0] += 1;
+  const int total = B * HW;
+  double s = 0.0, s2 = 0.0;
+  for (int i = threadIdx.x; i < total; i += blockDim.x) {
+    const unsigned b = d_hw.div((unsigned)i);
+    const unsigned r = d_hw.mod((unsigned)i, b);
+    const double v = (double)x[((long)b * C + c) * HW + r];
+    s += v;
+    s2 += v * v;
+  }
+  double ts = slk_block_sum(s, scratch);
+  __syncthreads();
+  double ts2 = slk_block_sum(s2, scratch);
+  if (threadIdx.x == 0) {
+    const float n = (float)total;
+    const float m = (float)ts / n;
+    const float v = fmaxf((float)ts2 / n - m * m, 0.f);
+    mean[c] = m;
+    invstd[c] = rsqrtf(v + eps);
+    if (running_mean != nullptr) {
+      const float unbiased = v * (n / fmaxf(n - 1.f, 1.f));
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    }
+  }
+}
+
 __global__ void bn_fwd_kernel(const float* __restrict__ x, float* __restrict__ y,
                               const float* __restrict__ mean,
                               const float* __restrict__ invstd,
@@ -138,6 +179,40 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ x,
     const long chunks = gridDim.y;
     slab[(long)blockIdx.y * C + c] = (float)ts;
     slab[chunks * C + (long)blockIdx.y * C + c] = (float)tsx;
+  }
+}
+
+// chunks==1 twin of bn_bwd_reduce: per-channel block writes the two sums
+// directly (no slab, no bn_bwd_finalize launch)
+__global__ void bn_bwd_reduce_one_kernel(const float* __restrict__ x,
+                                         const float* __restrict__ gy,
+                                         const float* __restrict__ relu_y,
+                                         const float* __restrict__ mean,
+                                         const float* __restrict__ invstd,
+                                         float* __restrict__ sum_gy,
+                                         float* __restrict__ sum_gy_xhat,
+                                         int B, int C, int HW, FastDiv d_hw) {
+  __shared__ double scratch[16];
+  const int c = blockIdx.x;
+  const float m = mean[c], is = invstd[c];
+  const int total = B * HW;
+  double s = 0.0, sx = 0.0;
+  for (int i = threadIdx.x; i < total; i += blockDim.x) {
+    const unsigned b = d_hw.div((unsigned)i);
+    const unsigned r = d_hw.mod((unsigned)i, b);
+    const long off = ((long)b * C + c) * HW + r;
+    float gf = gy[off];
+    if (relu_y != nullptr && relu_y[off] <= 0.f) gf = 0.f;
+    const double g = (double)gf;
+    s += g;
+    sx += g * (double)((x[off] - m) * is);
+  }
+  double ts = slk_block_sum(s, scratch);
+  __syncthreads();
+  double tsx = slk_block_sum(sx, scratch);
+  if (threadIdx.x == 0) {
+    sum_gy[c] = (float)ts;
+    sum_gy_xhat[c] = (float)tsx;
   }
 }
 
@@ -200,9 +275,22 @@ std::vector<at::Tensor> bn2d_stats_fused(const at::Tensor& x,
   auto invstd = at::empty({C}, x.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int chunks = bn_chunks((long)B * HW);
-  auto slab = at::empty({2, chunks, C}, x.options());
   FastDiv d_hw;
   d_hw.init(HW);
+  if (chunks == 1) {
+    hipLaunchKernelGGL(bn_stats_one_kernel, dim3(C), dim3(256), 0, stream,
+                       x.data_ptr<float>(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(),
+                       running_mean.has_value() ? running_mean->data_ptr<float>()
+                                                : nullptr,
+                       running_var.has_value() ? running_var->data_ptr<float>()
+                                               : nullptr,
+                       num_batches_tracked.has_value()
+                           ? num_batches_tracked->data_ptr<long>() : nullptr,
+                       B, C, HW, (float)momentum, (float)eps, d_hw);
+    return {mean, invstd};
+  }
+  auto slab = at::empty({2, chunks, C}, x.options());
   hipLaunchKernelGGL(bn_partial_kernel, dim3(C, chunks), dim3(256), 0, stream,
                      x.data_ptr<float>(), slab.data_ptr<float>(), B, C, HW,
                      d_hw);
@@ -252,9 +340,16 @@ static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tens
   auto gx = at::empty_like(x);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int rchunks = bn_chunks((long)B * HW);
-  auto slab = at::empty({2, rchunks, C}, x.options());
   FastDiv d_hw;
   d_hw.init(HW);
+  if (rchunks == 1) {
+    hipLaunchKernelGGL(bn_bwd_reduce_one_kernel, dim3(C), dim3(256), 0, stream,
+                       x.data_ptr<float>(), gy.data_ptr<float>(), ry,
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       sum_gy.data_ptr<float>(), sum_gy_xhat.data_ptr<float>(),
+                       B, C, HW, d_hw);
+  } else {
+  auto slab = at::empty({2, rchunks, C}, x.options());
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C, rchunks),
                      dim3(256), 0, stream,
                      x.data_ptr<float>(), gy.data_ptr<float>(), ry,
@@ -263,6 +358,7 @@ static std::vector<at::Tensor> bn2d_bwd_impl(const at::Tensor& x, const at::Tens
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
                      stream, slab.data_ptr<float>(), rchunks,
                      sum_gy.data_ptr<float>(), sum_gy_xhat.data_ptr<float>(), C);
+  }
   int grid = (int)std::min<long>((total + 255) / 256, 2048);
   FastDiv d_c;
   d_c.init(C);

@@ -167,9 +167,13 @@ class Conv2dFn(Function):
                 # conv feeding a TRAINING-mode BatchNorm: the bias gradient is
                 # analytically zero (sum over the batch of the BN-backward
                 # output is gamma*invstd*(sum_gy - sum_gy - sum_gy_xhat*
-                # sum(xhat)/n) and sum(xhat) == 0 by construction), so the
-                # two-kernel reduction is replaced by a constant
-                gb = gy.new_zeros(weight.shape[0])
+                # sum(xhat)/n) and sum(xhat) == 0 by construction).  Return
+                # None rather than materializing zeros: AccumulateGrad then
+                # skips the parameter entirely (no fill launch, no buffer),
+                # and the fused optimizers already skip None-grad params —
+                # with zero grad the momentum/Adam state stays zero, so the
+                # update is bit-identical to accumulating explicit zeros.
+                gb = None
             else:
                 gb = native().conv2d_bwd_bias(gy)
         return gx, gw, gb, None, None, None

@@ -13,6 +13,47 @@ import torch
 
 
 
+SLK_OPT_CHUNK = 16384
+
+
+def _build_desc(host_pin, live_params, live_grads, s1, s2, device):
+    """Fill the fused-step pointer table (layout = optim.hip make_opt_desc:
+    prefix[n+1], numels[n], param/grad/state1/state2 ptrs[n]) into the
+    preallocated PINNED host buffer and async-copy it to the device.
+
+    Built in Python (not via the pageable-H2D C++ helper) so it is legal
+    INSIDE hipGraph capture: the device tensor comes from the capture pool and
+    the H2D copy from pinned memory is recorded as a graph memcpy node that
+    re-reads the (kept-alive, constant) pinned buffer on every replay."""
+    n = len(live_params)
+    vals = [0] * (n + 1)
+    numels, pp, gp, b1, b2 = [], [], [], [], []
+    chunks = 0
+    for i, p in enumerate(live_params):
+        vals[i] = chunks
+        ne = p.numel()
+        numels.append(ne)
+        chunks += (ne + SLK_OPT_CHUNK - 1) // SLK_OPT_CHUNK
+        pp.append(p.data_ptr())
+        gp.append(live_grads[i].data_ptr())
+        b1.append(s1[i].data_ptr())
+        b2.append(s2[i].data_ptr())
+    vals[n] = chunks
+    vals += numels + pp + gp + b1 + b2
+    m = len(vals)
+    host_pin[:m].copy_(torch.tensor(vals, dtype=torch.int64))  # host-side
+    desc = torch.empty(m, dtype=torch.int64, device=device)
+    desc.copy_(host_pin[:m], non_blocking=True)
+    return desc, chunks
+
+
+def _pinned_host(n_params: int) -> torch.Tensor:
+    host = torch.empty(6 * n_params + 1, dtype=torch.int64)
+    if torch.cuda.is_available():
+        host = host.pin_memory()  # preallocated OUTSIDE any graph capture
+    return host
+
+
 class FusedSGD:
     def __init__(self, params, lr: float, momentum: float = 0.0,
                  weight_decay: float = 0.0):
@@ -23,25 +64,28 @@ class FusedSGD:
         self.bufs = [torch.zeros_like(p) for p in self.params]
         self.steps = 0
         self._desc_cache = None  # (ptr_signature, desc_tensor, n, chunks)
+        self._host = _pinned_host(len(self.params))
         # release_grads=True (default): drop p.grad after the fused step so the
-        # next backward MOVES fresh gradients (no accumulate-add kernels).
-        # Set False under hipGraph capture (stable pointers required); the
-        # update kernel then zeroes grads in place instead.
+        # next backward MOVES fresh gradients in (AccumulateGrad steals the
+        # producing kernel's output tensor — no accumulate-add kernel per
+        # parameter).  This also holds under hipGraph capture: grads are None
+        # when the captured step runs, the capture-pool grad buffers stay
+        # stable across replays, and every backward kernel fully overwrites
+        # its gradient output, so no inter-replay zeroing is needed.
         self.release_grads = True
 
     def _desc(self, live):
-        """Cached device descriptor for the single-launch fused step; grads
-        keep stable storage because the update kernel zeroes them in place, so
-        the table survives across steps (rebuilt if any pointer moves)."""
+        """Cached device descriptor for the single-launch fused step; with
+        release_grads the allocator usually hands the same blocks back, so
+        the signature (and the table) is stable across eager steps."""
         sig = tuple(p.grad.data_ptr() for p, _ in live) + \
               tuple(p.data_ptr() for p, _ in live)
         if self._desc_cache is None or self._desc_cache[0] != sig:
-            from ..ops import native
             params = [p for p, _ in live]
-            desc = native().make_opt_desc(params, [p.grad for p, _ in live],
-                                          [b for _, b in live],
-                                          [b for _, b in live])
-            chunks = sum((p.numel() + 16383) // 16384 for p in params)
+            bufs = [b for _, b in live]
+            desc, chunks = _build_desc(self._host, params,
+                                       [p.grad for p, _ in live], bufs, bufs,
+                                       params[0].device)
             self._desc_cache = (sig, desc, len(params), chunks)
         return self._desc_cache[1], self._desc_cache[2], self._desc_cache[3]
 
@@ -95,18 +139,19 @@ class FusedAdamW:
         self.v = [torch.zeros_like(p) for p in self.params]
         self.steps = 0
         self._desc_cache = None
+        self._host = _pinned_host(len(self.params))
         self.release_grads = True  # see FusedSGD
 
     def _desc(self, live):
         sig = tuple(p.grad.data_ptr() for p, _, _ in live) + \
               tuple(p.data_ptr() for p, _, _ in live)
         if self._desc_cache is None or self._desc_cache[0] != sig:
-            from ..ops import native
             params = [p for p, _, _ in live]
-            desc = native().make_opt_desc(params, [p.grad for p, _, _ in live],
-                                          [m for _, m, _ in live],
-                                          [v for _, _, v in live])
-            chunks = sum((p.numel() + 16383) // 16384 for p in params)
+            desc, chunks = _build_desc(self._host, params,
+                                       [p.grad for p, _, _ in live],
+                                       [m for _, m, _ in live],
+                                       [v for _, _, v in live],
+                                       params[0].device)
             self._desc_cache = (sig, desc, len(params), chunks)
         return self._desc_cache[1], self._desc_cache[2], self._desc_cache[3]
 

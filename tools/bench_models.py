@@ -57,16 +57,18 @@ def bench_model(model, data, cut, batch, steps=64, warmup=16, lora=False,
 
     graph = None
     if graphs:
-        # graph capture freezes grad/descriptor pointers: keep grads alive
-        # (the optimizers normally release them after the fused step)
-        o1.release_grads = False
-        o2.release_grads = False
+        # grads are None entering capture: AccumulateGrad steals the backward
+        # kernels' outputs (no per-param add_), and the optimizer descriptor
+        # is rebuilt inside capture via the pinned-memcpy path (see bench.py)
         sstream = torch.cuda.Stream()
         sstream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(sstream):
             for _ in range(3):
                 step()
         torch.cuda.current_stream().wait_stream(sstream)
+        # keep the stolen grad buffers alive (no mid-capture free; see bench.py)
+        o1.release_grads = False
+        o2.release_grads = False
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):
             step()
