@@ -318,11 +318,12 @@ constexpr int WF_VLD = WF_T + 1;    // V_lds[f][ci][t]
 // LINEAR in M, so slices transform their PARTIAL sums and atomicAdd into a
 // zeroed y — this fills the chip for small-T/high-Ci shapes whose natural
 // grid is far below one block per CU.  Slice 0 adds the bias.
-__global__ __launch_bounds__(256, 2) void wino_fused_kernel(
-    const float* __restrict__ x, const float* __restrict__ U,
+template <bool FLIP>
+__global__ __launch_bounds__(256, 3) void wino_fused_kernel(
+    const float* __restrict__ x, const float* __restrict__ w,
     const float* __restrict__ bias, float* __restrict__ y, int B, int Ci,
     int H, int W, int Co, int OH, int OW, int tH, int tW, int pad,
-    int ci_per, FastDiv d_thw, FastDiv d_tw) {
+    int ci_per, int CW, FastDiv d_thw, FastDiv d_tw) {
   const int ci_begin = blockIdx.z * ci_per;
   const int ci_end = min(Ci, ci_begin + ci_per);
   const bool accumulate = gridDim.z > 1;
@@ -360,14 +361,19 @@ __global__ __launch_bounds__(256, 2) void wino_fused_kernel(
   const int ih0 = (int)vth * 2 - pad;
   const int iw0 = (int)vtw * 2 - pad;
   const long xplane = ((long)vb * Ci + v_ci) * H * W;
-  // U: each thread stages 16 elements (256 threads x 16 = 16f x 32co x 8ci)
-  const int u_lin = tid;
+  // weights: thread -> one (co, ci) pair (32co x 8ci = 256); the 3x3 taps
+  // load raw and transform IN-KERNEL (U = G w G^T, the exact wino_wt math) —
+  // the separate wt launch and the 16*Co*Ci frequency tensor's HBM
+  // write+read round trip disappear.  FLIP (bwd-data) swaps the channel
+  // roles and rotates the taps 180 deg ((2-r, 2-s) == tap 8-k).
+  const int w_co = tid >> 3;
+  const int w_ci = tid & 7;
 
-  // software pipeline: next step's 4x4 patch + 16 U elements load into
-  // registers UNDER the (long, 32-MFMA) compute phase; the transform + LDS
+  // software pipeline: next step's 4x4 patch + 9 weight taps load into
+  // registers UNDER the (long, 32-MFMA) compute phase; the transforms + LDS
   // stores run at the top of the next iteration
   float d[4][4];
-  float ur[16];
+  float wr[9];
 
   auto load_regs = [&](int ci0) {
     const float* xp = x + xplane + (long)ci0 * H * W;
@@ -382,13 +388,11 @@ __global__ __launch_bounds__(256, 2) void wino_fused_kernel(
         d[a][bb] = v ? xp[(long)ih * W + iw] : 0.f;
       }
     }
+    const long wbase = FLIP
+        ? ((long)(ci0 + w_ci) * CW + (co0 + w_co)) * 9
+        : ((long)(co0 + w_co) * CW + (ci0 + w_ci)) * 9;
     #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      const int e = u_lin + j * 256;        // 0..4095
-      const int f = e >> 8;                 // 0..15
-      const int r = e & 255;                // co*8 + ci
-      ur[j] = U[((long)f * Co + co0 + (r >> 3)) * Ci + ci0 + (r & 7)];
-    }
+    for (int k = 0; k < 9; ++k) wr[k] = w[wbase + (FLIP ? 8 - k : k)];
   };
 
   auto store_stage = [&]() {
@@ -407,12 +411,25 @@ __global__ __launch_bounds__(256, 2) void wino_fused_kernel(
       Vl[((a * 4 + 2) * WF_CK + v_ci) * WF_VLD + v_t] = u[a][2] - u[a][1];
       Vl[((a * 4 + 3) * WF_CK + v_ci) * WF_VLD + v_t] = u[a][1] - u[a][3];
     }
+    // U = G w G^T for this thread's (co, ci) pair — same op order as
+    // wino_wt_kernel, so the result is bit-identical to the staged path
+    float t0[3], t1[3], t2[3], t3[3];
     #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      const int e = u_lin + j * 256;
-      const int f = e >> 8;
-      const int r = e & 255;
-      Ul[(f * WF_CK + (r & 7)) * WF_ULD + (r >> 3)] = ur[j];
+    for (int bb = 0; bb < 3; ++bb) {
+      const float g0 = wr[bb], g1 = wr[3 + bb], g2 = wr[6 + bb];
+      t0[bb] = g0;
+      t1[bb] = 0.5f * (g0 + g1 + g2);
+      t2[bb] = 0.5f * (g0 - g1 + g2);
+      t3[bb] = g2;
+    }
+    const float* tr[4] = {t0, t1, t2, t3};
+    #pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      const float* t = tr[a];
+      Ul[((a * 4 + 0) * WF_CK + w_ci) * WF_ULD + w_co] = t[0];
+      Ul[((a * 4 + 1) * WF_CK + w_ci) * WF_ULD + w_co] = 0.5f * (t[0] + t[1] + t[2]);
+      Ul[((a * 4 + 2) * WF_CK + w_ci) * WF_ULD + w_co] = 0.5f * (t[0] - t[1] + t[2]);
+      Ul[((a * 4 + 3) * WF_CK + w_ci) * WF_ULD + w_co] = t[2];
     }
   };
 
@@ -751,21 +768,9 @@ at::Tensor conv2d_wino_fused(const at::Tensor& x, const at::Tensor& w,
   TORCH_CHECK(Co % 32 == 0 && T % 32 == 0 && Ci % 8 == 0 && OH % 2 == 0
               && OW % 2 == 0, "wino_fused shape requirements");
   auto stream = c10::hip::getCurrentHIPStream().stream();
-
-  auto U = at::empty({16, Co, Ci}, x.options());
-  {
-    const long tot = (long)Co * Ci * 4;
-    const int grid = (int)std::min<long>((tot + 255) / 256, 4096);
-    if (flip) {
-      hipLaunchKernelGGL(wino_wt_kernel<true>, dim3(grid), dim3(256), 0,
-                         stream, wc.data_ptr<float>(), U.data_ptr<float>(),
-                         (int)w.size(0), (int)w.size(1));
-    } else {
-      hipLaunchKernelGGL(wino_wt_kernel<false>, dim3(grid), dim3(256), 0,
-                         stream, wc.data_ptr<float>(), U.data_ptr<float>(),
-                         (int)w.size(0), (int)w.size(1));
-    }
-  }
+  // the weight transform runs IN-KERNEL (one (co,ci) pair per thread per
+  // staging step) — no wt launch, no 16*Co*Ci frequency tensor
+  const int CW = (int)w.size(1);
   // ci-split so the grid reaches ~1 block/CU (output transform is linear in
   // M: slices atomicAdd partial y tiles; slice count keeps ci chunks at
   // multiples of the CK=8 step)
@@ -790,11 +795,19 @@ at::Tensor conv2d_wino_fused(const at::Tensor& x, const at::Tensor& w,
   d_thw.init(tH * tW);
   d_tw.init(tW);
   dim3 grid(T / 32, Co / 32, splits);
-  hipLaunchKernelGGL(wino_fused_kernel, grid, dim3(256), 0, stream,
-                     xc.data_ptr<float>(), U.data_ptr<float>(),
-                     bias.has_value() ? bias->data_ptr<float>() : nullptr,
-                     y.data_ptr<float>(), B, Ci, H, W, Co, OH, OW, tH, tW,
-                     pad, ci_per, d_thw, d_tw);
+  if (flip) {
+    hipLaunchKernelGGL(wino_fused_kernel<true>, grid, dim3(256), 0, stream,
+                       xc.data_ptr<float>(), wc.data_ptr<float>(),
+                       bias.has_value() ? bias->data_ptr<float>() : nullptr,
+                       y.data_ptr<float>(), B, Ci, H, W, Co, OH, OW, tH, tW,
+                       pad, ci_per, CW, d_thw, d_tw);
+  } else {
+    hipLaunchKernelGGL(wino_fused_kernel<false>, grid, dim3(256), 0, stream,
+                       xc.data_ptr<float>(), wc.data_ptr<float>(),
+                       bias.has_value() ? bias->data_ptr<float>() : nullptr,
+                       y.data_ptr<float>(), B, Ci, H, W, Co, OH, OW, tH, tW,
+                       pad, ci_per, CW, d_thw, d_tw);
+  }
   return y;
 }
 
