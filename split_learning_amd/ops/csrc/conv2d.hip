@@ -69,6 +69,7 @@ static inline bool wino_fused_ok(int Cin, int Cout, int T, int OH, int OW) {
   if (v == 0) return false;
   if ((OH | OW) & 1) return false;
   if (Cout % 32 != 0 || T % 32 != 0 || Cin % 8 != 0) return false;
+  if (v >= 3) return true;  // force: ci-split fills the grid (sweeps)
   return (long)(T / 32) * (Cout / 32) >= 256;
 }
 
@@ -77,6 +78,14 @@ static inline bool wino_fused_ok(int Cin, int Cout, int T, int OH, int OW) {
 // at either end of the stack (64ch/32x32 and >=256ch tails); the 128ch
 // middle loses to V/M HBM inflation (tools/wino_check.py table in
 // profiles/SUMMARY.md).
+static inline bool wino_env_on() {
+  static int v = [] {
+    const char* e = std::getenv("SLK_WINO");
+    return e ? atoi(e) : 1;
+  }();
+  return v != 0;
+}
+
 static inline bool wino_wins_bwdw(int Ci, int Co, int H, int OH, int OW) {
   static int v = [] {
     const char* e = std::getenv("SLK_WINO");
@@ -650,7 +659,18 @@ at::Tensor conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w, int stride
   // gx = winograd-conv of gy with rotated/transposed weights at pad' = 2-p
   if (stride == 1 && KH == 3 && KW == 3 && pad <= 2) {
     const int Tw = B * (H / 2) * (W / 2);
-    if (H % 2 == 0 && W % 2 == 0 && wino_fused_ok(Co, Ci, Tw, H, W)) {
+    bool fused = H % 2 == 0 && W % 2 == 0 && wino_fused_ok(Co, Ci, Tw, H, W);
+    // backward-data ONLY: the ci-split fused kernel also wins on the
+    // small-T high-channel tails (measured, 40-iter A/B under SLK_WINO=3:
+    // 256ch 8x8 44.6 vs 49.3 us, 256->512 4x4 31.6 vs 38.1, 512ch 4x4 45.3
+    // vs 49.9; the 2x2 layers lose 62 vs 35 and stay out).  Forward at the
+    // same shapes measured WORSE fused and keeps the >=256-block rule.
+    if (!fused && H % 2 == 0 && W % 2 == 0 && H >= 4 && Ci >= 256
+        && Ci % 32 == 0 && Co % 8 == 0 && Tw % 32 == 0
+        && (long)(Tw / 32) * (Ci / 32) >= 32 && wino_env_on()) {
+      fused = true;  // Ci >= 256 keeps this to the measured shapes only
+    }
+    if (fused) {
       return conv2d_wino_fused(gy, w, c10::nullopt, 2 - pad, /*flip=*/true);
     }
     if (wino_wins(Co, Ci, gy.size(2), 3, 3, 1, H, W)) {
