@@ -297,3 +297,25 @@ def test_param_broadcast_roundtrip(tmp_path):
     port = _free_port()
     mp.spawn(_bcast_unit_worker, args=(2, port, str(tmp_path)), nprocs=2,
              join=True)
+
+
+def test_opt_desc_python_matches_cpp():
+    """The fused-step pointer table is built in Python (pinned host buffer +
+    async copy, legal inside hipGraph capture — parallel/optim._build_desc);
+    its layout must stay bit-identical to the C++ builder the kernels were
+    written against (ops/csrc/optim.hip make_opt_desc: prefix[n+1], numel[n],
+    param/grad/state1/state2 ptrs[n])."""
+    import torch
+
+    from split_learning_amd.ops import native
+    from split_learning_amd.parallel.optim import _build_desc, _pinned_host
+
+    p = [torch.randn(7), torch.randn(3, 5), torch.randn(20000)]
+    g = [torch.randn_like(t) for t in p]
+    b = [torch.zeros_like(t) for t in p]
+    ref = native().make_opt_desc(p, g, b, b)
+    host = _pinned_host(len(p))
+    got, chunks = _build_desc(host, p, g, b, b, torch.device("cpu"))
+    assert torch.equal(ref, got)
+    # chunks = ceil(numel / 16384) summed; the 20000-element tensor needs 2
+    assert chunks == 1 + 1 + 2
