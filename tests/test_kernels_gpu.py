@@ -197,10 +197,14 @@ def test_conv2d_bwd(B, Ci, H, W, Co, K, s, p):
 
 # ---------------- BatchNorm ----------------
 
-def test_bn2d_train_fwd_bwd():
+@pytest.mark.parametrize("B,H", [(8, 16), (32, 32)])
+def test_bn2d_train_fwd_bwd(B, H):
+    """(8,16): B*HW=2048 -> the single-launch bn_stats_one/bwd_reduce_one
+    kernels (chunks==1); (32,32): B*HW=32768 -> the chunked partial+finalize
+    path.  Both must match torch."""
     from split_learning_amd.ops import functional as hf
     torch.manual_seed(0)
-    x = torch.randn(8, 32, 16, 16, device="cuda")
+    x = torch.randn(B, 32, H, H, device="cuda")
     gamma = torch.randn(32, device="cuda", requires_grad=True)
     beta = torch.randn(32, device="cuda", requires_grad=True)
     rm = torch.zeros(32, device="cuda")
