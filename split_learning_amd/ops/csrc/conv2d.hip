@@ -57,10 +57,12 @@ at::Tensor conv2d_wino_fused(const at::Tensor&, const at::Tensor&,
 at::Tensor conv2d_wino_bwdw_fused(const at::Tensor&, const at::Tensor&, int);
 
 // fused-Winograd eligibility: shape constraints + grid fill.  The fused
-// kernel's grid is (T/32) x (Cout/32) with no ci-split, and the measured
-// win/lose line falls EXACTLY at >= 256 blocks (1 per CU): above it the
-// fused kernel beats every alternative (conv1_2 fwd 32.8 vs 44.3 us),
-// below it the chip underfills and it loses 2-4x (profiles/SUMMARY.md).
+// kernel's base grid is (T/32) x (Cout/32); for FORWARD the measured
+// win/lose line falls at >= 256 base blocks (1 per CU) — below it the
+// ci-split can fill the grid but the forward still measured WORSE fused
+// (8x8 tails: 52.4 vs 49.9 us), so forward keeps the 256 rule.
+// BACKWARD-DATA additionally wins on the ci-split small-T tails and gets
+// its own extension at the conv2d_bwd_data call site (profiles/SUMMARY.md).
 static inline bool wino_fused_ok(int Cin, int Cout, int T, int OH, int OW) {
   static int v = [] {
     const char* e = std::getenv("SLK_WINO");

@@ -54,6 +54,28 @@ def _pinned_host(n_params: int) -> torch.Tensor:
     return host
 
 
+# A captured graph's memcpy node re-reads its pinned source buffer on every
+# replay, so a buffer referenced by a live graph must never be overwritten or
+# freed.  Rebuilds outside capture therefore take a FRESH pinned buffer and
+# retire the old one here (tiny: <= a few hundred int64 each); the
+# preallocated per-optimizer buffer is reserved for the one build that can
+# happen INSIDE capture, where pinning anew would not be capture-legal.
+_RETIRED_HOSTS: List[torch.Tensor] = []
+
+
+def _host_for_build(opt) -> torch.Tensor:
+    """Pinned buffer for a (re)build: inside capture reuse the preallocated
+    one (pinning anew is not capture-legal); outside capture take a fresh
+    buffer and retire the old so any live graph's memcpy source stays
+    intact (see _RETIRED_HOSTS)."""
+    if torch.cuda.is_available() and torch.cuda.is_current_stream_capturing():
+        return opt._host
+    if opt._desc_cache is not None:
+        _RETIRED_HOSTS.append(opt._host)
+        opt._host = _pinned_host(len(opt.params))
+    return opt._host
+
+
 class FusedSGD:
     def __init__(self, params, lr: float, momentum: float = 0.0,
                  weight_decay: float = 0.0):
@@ -83,7 +105,7 @@ class FusedSGD:
         if self._desc_cache is None or self._desc_cache[0] != sig:
             params = [p for p, _ in live]
             bufs = [b for _, b in live]
-            desc, chunks = _build_desc(self._host, params,
+            desc, chunks = _build_desc(_host_for_build(self), params,
                                        [p.grad for p, _ in live], bufs, bufs,
                                        params[0].device)
             self._desc_cache = (sig, desc, len(params), chunks)
@@ -147,7 +169,7 @@ class FusedAdamW:
               tuple(p.data_ptr() for p, _, _ in live)
         if self._desc_cache is None or self._desc_cache[0] != sig:
             params = [p for p, _, _ in live]
-            desc, chunks = _build_desc(self._host, params,
+            desc, chunks = _build_desc(_host_for_build(self), params,
                                        [p.grad for p, _, _ in live],
                                        [m for _, m, _ in live],
                                        [v for _, _, v in live],

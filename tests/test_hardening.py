@@ -319,3 +319,23 @@ def test_opt_desc_python_matches_cpp():
     assert torch.equal(ref, got)
     # chunks = ceil(numel / 16384) summed; the 20000-element tensor needs 2
     assert chunks == 1 + 1 + 2
+
+
+def test_opt_desc_rebuild_retires_host_buffer():
+    """A captured graph's memcpy node re-reads its pinned source on every
+    replay, so a descriptor REBUILD (outside capture) must take a fresh host
+    buffer and retire the old one rather than overwrite it in place."""
+    import torch
+
+    from split_learning_amd.parallel import optim as O
+
+    p = [torch.randn(100, requires_grad=True)]
+    o = O.FusedSGD(p, lr=0.1, momentum=0.9)
+    p[0].grad = torch.randn(100)
+    live = list(zip(o.params, o.bufs))
+    h0 = o._host
+    o._desc(live)
+    p[0].grad = torch.randn(100)  # new storage -> signature miss -> rebuild
+    o._desc(live)
+    assert o._host is not h0
+    assert O._RETIRED_HOSTS and O._RETIRED_HOSTS[-1] is h0
